@@ -1,0 +1,141 @@
+// Fused residual + LayerNorm / RMSNorm kernels (bf16 I/O, fp32 math).
+//
+// Replaces the LayerNorm/RMSNorm work the reference delegated to HF
+// transformers model forwards (reference: distllm/embed/encoders/auto.py
+// encode() running AutoModel).
+//
+// Design (guide: Appendix B element-wise/reduction rules):
+//  - one WAVE per row (row = hidden vector of size H), 4 rows per 256-thread
+//    block; bf16 loads vectorized as s16x8 (16 B/lane);
+//  - fp32 accumulate, single pass, row held in registers. Per-lane chunk
+//    count is a template parameter so every register index is
+//    compile-time (runtime-indexed ext_vector arrays spill to scratch).
+//  - optional residual add fused, optional write-back of the summed
+//    residual stream (pre-norm architectures need it for the next skip).
+#include "common.h"
+
+template <int NCL, bool HAS_RES, bool WRITE_SUM, bool RMS>
+__global__ void __launch_bounds__(256)
+norm_fwd_kernel(const unsigned short* __restrict__ x,
+                const unsigned short* __restrict__ res,
+                const unsigned short* __restrict__ w,
+                const unsigned short* __restrict__ b,
+                unsigned short* __restrict__ out,
+                unsigned short* __restrict__ sum_out,
+                int n_rows, int H, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int waves_per_grid = (gridDim.x * blockDim.x) >> 6;
+  const int chunks = H >> 3;  // H multiple of 8 enforced host-side
+
+  for (int row = blockIdx.x * 4 + wid; row < n_rows; row += waves_per_grid) {
+    const unsigned short* xr = x + (long)row * H;
+    const unsigned short* rr = HAS_RES ? res + (long)row * H : nullptr;
+
+    float v[NCL][8];
+    float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+    for (int k = 0; k < NCL; ++k) {
+      const int c = lane + k * WAVE;
+      if (c < chunks) {
+        s16x8 xv = *reinterpret_cast<const s16x8*>(xr + c * 8);
+        float f[8];
+        bf8_to_f32(xv, f);
+        if (HAS_RES) {
+          s16x8 rv = *reinterpret_cast<const s16x8*>(rr + c * 8);
+          float g[8];
+          bf8_to_f32(rv, g);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) f[i] += g[i];
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          v[k][i] = f[i];
+          sum += f[i];
+          sumsq += f[i] * f[i];
+        }
+        if (WRITE_SUM) {
+          s16x8 sv;
+#pragma unroll
+          for (int i = 0; i < 8; ++i) sv[i] = (short)f2bf(f[i]);
+          *reinterpret_cast<s16x8*>(sum_out + (long)row * H + c * 8) = sv;
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[k][i] = 0.f;
+      }
+    }
+    sum = wave_reduce_sum(sum);
+    sumsq = wave_reduce_sum(sumsq);
+    float mean, inv;
+    if (RMS) {
+      mean = 0.f;
+      inv = rsqrtf(sumsq / H + eps);
+    } else {
+      mean = sum / H;
+      inv = rsqrtf(fmaxf(sumsq / H - mean * mean, 0.f) + eps);
+    }
+#pragma unroll
+    for (int k = 0; k < NCL; ++k) {
+      const int c = lane + k * WAVE;
+      if (c < chunks) {
+        s16x8 wv = *reinterpret_cast<const s16x8*>(w + c * 8);
+        float wf[8];
+        bf8_to_f32(wv, wf);
+        float bf[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (b != nullptr) {
+          s16x8 bv = *reinterpret_cast<const s16x8*>(b + c * 8);
+          bf8_to_f32(bv, bf);
+        }
+        s16x8 ov;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float y = (v[k][i] - mean) * inv * wf[i] + bf[i];
+          ov[i] = (short)f2bf(y);
+        }
+        *reinterpret_cast<s16x8*>(out + (long)row * H + c * 8) = ov;
+      }
+    }
+  }
+}
+
+extern "C" void launch_norm_fwd(const void* x, const void* res, const void* w,
+                                const void* b, void* out, void* sum_out,
+                                int n_rows, int H, float eps, bool rms,
+                                hipStream_t stream) {
+  int blocks = min((n_rows + 3) / 4, 2048);
+  bool has_res = res != nullptr;
+  bool write_sum = sum_out != nullptr;
+  int ncl = (H / 8 + WAVE - 1) / WAVE;
+
+#define DISPATCH3(NCL, HR, WS, RM)                                             \
+  hipLaunchKernelGGL((norm_fwd_kernel<NCL, HR, WS, RM>), dim3(blocks),         \
+                     dim3(256), 0, stream, (const unsigned short*)x,           \
+                     (const unsigned short*)res, (const unsigned short*)w,     \
+                     (const unsigned short*)b, (unsigned short*)out,           \
+                     (unsigned short*)sum_out, n_rows, H, eps)
+#define DISPATCH_FLAGS(NCL)                                                    \
+  do {                                                                         \
+    if (rms) {                                                                 \
+      if (has_res && write_sum) DISPATCH3(NCL, true, true, true);              \
+      else if (has_res) DISPATCH3(NCL, true, false, true);                     \
+      else DISPATCH3(NCL, false, false, true);                                 \
+    } else {                                                                   \
+      if (has_res && write_sum) DISPATCH3(NCL, true, true, false);             \
+      else if (has_res) DISPATCH3(NCL, true, false, false);                    \
+      else DISPATCH3(NCL, false, false, false);                                \
+    }                                                                          \
+  } while (0)
+
+  switch (ncl) {
+    case 1: DISPATCH_FLAGS(1); break;
+    case 2: DISPATCH_FLAGS(2); break;
+    case 3: DISPATCH_FLAGS(3); break;
+    case 4: DISPATCH_FLAGS(4); break;
+    case 5: case 6: DISPATCH_FLAGS(6); break;
+    case 7: case 8: DISPATCH_FLAGS(8); break;
+    default: DISPATCH_FLAGS(16); break;  // up to H=8192
+  }
+#undef DISPATCH_FLAGS
+#undef DISPATCH3
+}

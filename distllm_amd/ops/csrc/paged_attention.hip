@@ -1,0 +1,191 @@
+// Paged-KV decode attention (single new token per sequence), bf16.
+// Replaces vLLM's paged attention (reference delegated generation to
+// vLLM, generate/generators/vllm_backend.py).
+//
+// Cache layout: kcache/vcache [nblocks, HKV, BLK, D] bf16, BLK=16.
+// block_table [B, max_blocks] int32 ; ctx_len [B] int32 (tokens in ctx,
+// including the one being attended as the newest position).
+// q: [B, H, D] bf16 ; out: [B, H, D] bf16.
+//
+// Decode is KV-bandwidth-bound: one workgroup per (b, h); 4 waves split
+// the context; lane-per-key dot for QK, d-sliced coalesced accumulate
+// for PV through an LDS p-buffer; cross-wave combine via LDS (m, l, acc).
+#include "common.h"
+
+#define PBLK 16  // tokens per KV block
+
+template <int D>
+__global__ void __launch_bounds__(256)
+paged_attn_kernel(const unsigned short* __restrict__ q,
+                  const unsigned short* __restrict__ kcache,
+                  const unsigned short* __restrict__ vcache,
+                  const int* __restrict__ block_table,
+                  const int* __restrict__ ctx_len, unsigned short* __restrict__ out,
+                  int B, int H, int HKV, int max_blocks, float scale) {
+  const int b = blockIdx.x / H;
+  const int h = blockIdx.x % H;
+  const int kvh = h / (H / HKV);
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int L = ctx_len[b];
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: p_lds [4][64] f32 | q_lds [D] f32 | wave state m,l [4][2]
+  float* p_lds = reinterpret_cast<float*>(smem);                 // 1 KiB
+  float* q_lds = p_lds + 4 * 64;                                 // D f32
+  float* wstate = q_lds + D;                                     // 8 f32
+  float* wacc = wstate + 8;                                      // 4*D f32
+
+  // load q into LDS as f32 (scaled)
+  for (int d = threadIdx.x; d < D; d += 256)
+    q_lds[d] = bf2f(q[((long)b * H + h) * D + d]) * scale;
+  __syncthreads();
+
+  // each wave processes keys [wid*64 + 256*t ...] lane-per-key
+  float m = -1e30f, l = 0.f;
+  float acc[2] = {0.f, 0.f};  // d-slice accumulator: lane owns D/64 dims
+  constexpr int DPL = D / 64; // 1 (D=64) or 2 (D=128)
+
+  const long blk_stride = (long)HKV * PBLK * D;
+
+  for (int t0 = wid * 64; t0 < L; t0 += 256) {
+    const int key = t0 + lane;
+    float s = -INFINITY;
+    if (key < L) {
+      const int blk = block_table[(long)b * max_blocks + key / PBLK];
+      const unsigned short* krow =
+          kcache + (long)blk * blk_stride + ((long)kvh * PBLK + key % PBLK) * D;
+      float dot = 0.f;
+#pragma unroll
+      for (int c = 0; c < D / 8; ++c) {
+        s16x8 kv8 = *reinterpret_cast<const s16x8*>(krow + c * 8);
+        float kf[8];
+        bf8_to_f32(kv8, kf);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) dot += kf[i] * q_lds[c * 8 + i];
+      }
+      s = dot;  // q already scaled
+    }
+    // wave online softmax
+    float pmax = wave_reduce_max(s);
+    float mn = fmaxf(fmaxf(m, pmax), -1e30f);
+    float alpha = exp2f((m - mn) * 1.4426950408889634f);
+    float p = (key < L) ? exp2f((s - mn) * 1.4426950408889634f) : 0.f;
+    float rs = wave_reduce_sum(p);
+    l = l * alpha + rs;
+    m = mn;
+    p_lds[wid * 64 + lane] = p;
+    // rescale accumulator
+#pragma unroll
+    for (int i = 0; i < DPL; ++i) acc[i] *= alpha;
+    // PV: lane owns dims d = lane*DPL + i ... use coalesced V reads:
+    // iterate the 64 keys of this wave's tile; lane reads V[key][d_own].
+    // (wave-internal LDS p broadcast; no cross-wave sync needed)
+#pragma unroll 4
+    for (int kk = 0; kk < 64; ++kk) {
+      const int key2 = t0 + kk;
+      if (key2 >= L) break;
+      const float pv = p_lds[wid * 64 + kk];
+      if (pv == 0.f) continue;
+      const int blk = block_table[(long)b * max_blocks + key2 / PBLK];
+      const unsigned short* vrow =
+          vcache + (long)blk * blk_stride + ((long)kvh * PBLK + key2 % PBLK) * D;
+#pragma unroll
+      for (int i = 0; i < DPL; ++i)
+        acc[i] += pv * bf2f(vrow[lane * DPL + i]);
+    }
+  }
+
+  // cross-wave combine through LDS
+  if (lane == 0) {
+    wstate[wid * 2] = m;
+    wstate[wid * 2 + 1] = l;
+  }
+#pragma unroll
+  for (int i = 0; i < DPL; ++i) wacc[wid * D + lane * DPL + i] = acc[i];
+  __syncthreads();
+
+  if (wid == 0) {
+    float M = -1e30f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) M = fmaxf(M, wstate[w * 2]);
+    float Ltot = 0.f;
+    float o[2] = {0.f, 0.f};
+#pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      const float a = exp2f((wstate[w * 2] - M) * 1.4426950408889634f);
+      Ltot += wstate[w * 2 + 1] * a;
+#pragma unroll
+      for (int i = 0; i < DPL; ++i) o[i] += a * wacc[w * D + lane * DPL + i];
+    }
+    const float inv = (Ltot > 0.f) ? 1.f / Ltot : 0.f;
+#pragma unroll
+    for (int i = 0; i < DPL; ++i)
+      out[((long)b * H + h) * D + lane * DPL + i] = f2bf(o[i] * inv);
+  }
+}
+
+extern "C" void launch_paged_attn(const void* q, const void* kcache,
+                                  const void* vcache, const int* block_table,
+                                  const int* ctx_len, void* out, int B, int H,
+                                  int HKV, int D, int max_blocks, float scale,
+                                  hipStream_t stream) {
+  const int lds = (4 * 64 + D + 8 + 4 * D) * 4;
+  dim3 grid(B * H);
+  if (D == 64)
+    hipLaunchKernelGGL((paged_attn_kernel<64>), grid, dim3(256), lds, stream,
+                       (const unsigned short*)q, (const unsigned short*)kcache,
+                       (const unsigned short*)vcache, block_table, ctx_len,
+                       (unsigned short*)out, B, H, HKV, max_blocks, scale);
+  else if (D == 128)
+    hipLaunchKernelGGL((paged_attn_kernel<128>), grid, dim3(256), lds, stream,
+                       (const unsigned short*)q, (const unsigned short*)kcache,
+                       (const unsigned short*)vcache, block_table, ctx_len,
+                       (unsigned short*)out, B, H, HKV, max_blocks, scale);
+}
+
+// Write one new token's K/V into the paged cache (decode step).
+// knew/vnew: [B, HKV, D] bf16; slot[b] = ctx position of the new token.
+template <int D>
+__global__ void __launch_bounds__(256)
+cache_put_kernel(const unsigned short* __restrict__ knew,
+                 const unsigned short* __restrict__ vnew,
+                 unsigned short* __restrict__ kcache,
+                 unsigned short* __restrict__ vcache,
+                 const int* __restrict__ block_table,
+                 const int* __restrict__ slot, int B, int HKV,
+                 int max_blocks) {
+  const long total = (long)B * HKV * D;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long blk_stride = (long)HKV * PBLK * D;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int d = (int)(i % D);
+    const int kvh = (int)((i / D) % HKV);
+    const int b = (int)(i / ((long)D * HKV));
+    const int pos = slot[b];
+    const int blk = block_table[(long)b * max_blocks + pos / PBLK];
+    const long off =
+        (long)blk * blk_stride + ((long)kvh * PBLK + pos % PBLK) * D + d;
+    kcache[off] = knew[i];
+    vcache[off] = vnew[i];
+  }
+}
+
+extern "C" void launch_cache_put(const void* knew, const void* vnew,
+                                 void* kcache, void* vcache,
+                                 const int* block_table, const int* slot,
+                                 int B, int HKV, int D, int max_blocks,
+                                 hipStream_t stream) {
+  const long total = (long)B * HKV * D;
+  int blocks = (int)min((total + 255) / 256, (long)2048);
+#define CP(DD)                                                                 \
+  hipLaunchKernelGGL((cache_put_kernel<DD>), dim3(blocks), dim3(256), 0,       \
+                     stream, (const unsigned short*)knew,                      \
+                     (const unsigned short*)vnew, (unsigned short*)kcache,     \
+                     (unsigned short*)vcache, block_table, slot, B, HKV,       \
+                     max_blocks)
+  if (D == 64) CP(64);
+  else CP(128);
+#undef CP
+}
