@@ -1,0 +1,121 @@
+"""Flagship benchmark: PubMedBERT bs512 embedding throughput (seqs/sec).
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`.
+For N>1 the driver launches this under torch.distributed.run with one
+rank per GPU (RCCL over xGMI). One timed step = embed one batch of
+`--batch-size` synthetic sequences of length `--seq-len` through the
+native HIP-kernel encoder forward + fused mean-pool + L2 normalize +
+async D2H into a pinned host buffer — the hot loop of the reference's
+headline config (examples/embed/AMP.nougat.pubmedbert.yaml:
+batch_size=512, maxlen 512, mean pool).
+
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch-size", type=int, default=512)
+    ap.add_argument("--seq-len", type=int, default=512)
+    ap.add_argument("--model", type=str, default="pubmedbert-base")
+    args = ap.parse_args()
+
+    from distllm_amd.models import build_model
+    from distllm_amd.parallel import (barrier, get_rank, get_world_size,
+                                      init_distributed)
+    from distllm_amd import ops
+
+    rank, world = init_distributed()
+    have_gpu = torch.cuda.is_available()
+    if have_gpu:
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+        dtype = torch.bfloat16
+    else:
+        device, dtype = "cpu", torch.float32
+
+    torch.manual_seed(1234 + rank)
+    model = build_model(args.model, device=device, dtype=dtype)
+    dims = model.dims
+
+    B, S = args.batch_size, args.seq_len
+    ids = torch.randint(4, dims.vocab_size, (B, S), device=device)
+    mask = torch.ones(B, S, device=device, dtype=torch.int32)
+    host = torch.empty(B, dims.hidden_size, dtype=torch.float32,
+                       pin_memory=have_gpu)
+    # pinned-host staged inputs: each step re-uploads (H2D in the loop,
+    # like the real dataloader path)
+    ids_host = ids.to("cpu").pin_memory() if have_gpu else ids
+    mask_host = mask.to("cpu").pin_memory() if have_gpu else mask
+
+    @torch.no_grad()
+    def step() -> None:
+        d_ids = ids_host.to(device, non_blocking=True)
+        d_mask = mask_host.to(device, non_blocking=True)
+        hidden = model(d_ids, d_mask)
+        pooled = ops.mean_pool(hidden, d_mask, normalize=True)
+        host.copy_(pooled, non_blocking=True)
+
+    for _ in range(args.warmup):
+        step()
+    barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if have_gpu:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks -> whole-job throughput
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = world * B * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "seqs/sec embedded (PubMedBERT bs512)",
+            "value": round(value, 2),
+            "unit": "seqs/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if have_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": world * B,
+                "seq_len": S,
+                "parallelism": f"dp{world}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

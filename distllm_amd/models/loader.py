@@ -1,0 +1,133 @@
+"""Local checkpoint loading (safetensors) with HF name mapping.
+
+No network in this environment: weights load only from local dirs. The
+mapping covers HF BertModel, EsmModel and Llama/Mistral layouts so a
+user with a downloaded checkpoint gets real weights; otherwise models
+run random-init (the benchmarks declare synthetic data/weights).
+"""
+
+from __future__ import annotations
+
+from pathlib import Path
+
+import torch
+
+
+def _load_safetensors_dir(path: Path) -> dict[str, torch.Tensor]:
+    from safetensors.torch import load_file
+
+    files = sorted(Path(path).glob("*.safetensors"))
+    if not files:
+        raise FileNotFoundError(f"no .safetensors under {path}")
+    state: dict[str, torch.Tensor] = {}
+    for f in files:
+        state.update(load_file(str(f)))
+    return state
+
+
+def _map_bert(src: dict, model) -> dict:
+    out = {}
+    p = "bert." if any(k.startswith("bert.") for k in src) else ""
+    out["word_emb"] = src[f"{p}embeddings.word_embeddings.weight"]
+    out["pos_emb"] = src[f"{p}embeddings.position_embeddings.weight"]
+    out["type_emb"] = src[f"{p}embeddings.token_type_embeddings.weight"]
+    out["emb_ln_w"] = src[f"{p}embeddings.LayerNorm.weight"]
+    out["emb_ln_b"] = src[f"{p}embeddings.LayerNorm.bias"]
+    for i in range(len(model.layers)):
+        lp = f"{p}encoder.layer.{i}."
+        q_w = src[lp + "attention.self.query.weight"]
+        k_w = src[lp + "attention.self.key.weight"]
+        v_w = src[lp + "attention.self.value.weight"]
+        out[f"layers.{i}.wqkv"] = torch.cat([q_w, k_w, v_w], 0)
+        out[f"layers.{i}.bqkv"] = torch.cat(
+            [src[lp + "attention.self.query.bias"],
+             src[lp + "attention.self.key.bias"],
+             src[lp + "attention.self.value.bias"]], 0)
+        out[f"layers.{i}.wo"] = src[lp + "attention.output.dense.weight"]
+        out[f"layers.{i}.bo"] = src[lp + "attention.output.dense.bias"]
+        out[f"layers.{i}.ln1_w"] = src[lp + "attention.output.LayerNorm.weight"]
+        out[f"layers.{i}.ln1_b"] = src[lp + "attention.output.LayerNorm.bias"]
+        out[f"layers.{i}.w1"] = src[lp + "intermediate.dense.weight"]
+        out[f"layers.{i}.b1"] = src[lp + "intermediate.dense.bias"]
+        out[f"layers.{i}.w2"] = src[lp + "output.dense.weight"]
+        out[f"layers.{i}.b2"] = src[lp + "output.dense.bias"]
+        out[f"layers.{i}.ln2_w"] = src[lp + "output.LayerNorm.weight"]
+        out[f"layers.{i}.ln2_b"] = src[lp + "output.LayerNorm.bias"]
+    return out
+
+
+def _map_llama(src: dict, model) -> dict:
+    out = {}
+    p = "model." if any(k.startswith("model.") for k in src) else ""
+    out["tok_emb"] = src[f"{p}embed_tokens.weight"]
+    out["final_norm"] = src[f"{p}norm.weight"]
+    if getattr(model, "lm_head", None) is not None and "lm_head.weight" in src:
+        out["lm_head"] = src["lm_head.weight"]
+    for i in range(len(model.layers)):
+        lp = f"{p}layers.{i}."
+        out[f"layers.{i}.wq"] = src[lp + "self_attn.q_proj.weight"]
+        out[f"layers.{i}.wk"] = src[lp + "self_attn.k_proj.weight"]
+        out[f"layers.{i}.wv"] = src[lp + "self_attn.v_proj.weight"]
+        out[f"layers.{i}.wo"] = src[lp + "self_attn.o_proj.weight"]
+        out[f"layers.{i}.in_norm"] = src[lp + "input_layernorm.weight"]
+        out[f"layers.{i}.post_norm"] = src[
+            lp + "post_attention_layernorm.weight"]
+        out[f"layers.{i}.wgate"] = src[lp + "mlp.gate_proj.weight"]
+        out[f"layers.{i}.wup"] = src[lp + "mlp.up_proj.weight"]
+        out[f"layers.{i}.wdown"] = src[lp + "mlp.down_proj.weight"]
+    return out
+
+
+def _map_esm2(src: dict, model) -> dict:
+    out = {}
+    p = "esm." if any(k.startswith("esm.") for k in src) else ""
+    out["word_emb"] = src[f"{p}embeddings.word_embeddings.weight"]
+    out["final_ln_w"] = src[f"{p}encoder.emb_layer_norm_after.weight"]
+    out["final_ln_b"] = src[f"{p}encoder.emb_layer_norm_after.bias"]
+    for i in range(len(model.layers)):
+        lp = f"{p}encoder.layer.{i}."
+        out[f"layers.{i}.wqkv"] = torch.cat(
+            [src[lp + "attention.self.query.weight"],
+             src[lp + "attention.self.key.weight"],
+             src[lp + "attention.self.value.weight"]], 0)
+        out[f"layers.{i}.bqkv"] = torch.cat(
+            [src[lp + "attention.self.query.bias"],
+             src[lp + "attention.self.key.bias"],
+             src[lp + "attention.self.value.bias"]], 0)
+        out[f"layers.{i}.wo"] = src[lp + "attention.output.dense.weight"]
+        out[f"layers.{i}.bo"] = src[lp + "attention.output.dense.bias"]
+        out[f"layers.{i}.ln1_w"] = src[lp + "attention.LayerNorm.weight"]
+        out[f"layers.{i}.ln1_b"] = src[lp + "attention.LayerNorm.bias"]
+        out[f"layers.{i}.ln2_w"] = src[lp + "LayerNorm.weight"]
+        out[f"layers.{i}.ln2_b"] = src[lp + "LayerNorm.bias"]
+        out[f"layers.{i}.w1"] = src[lp + "intermediate.dense.weight"]
+        out[f"layers.{i}.b1"] = src[lp + "intermediate.dense.bias"]
+        out[f"layers.{i}.w2"] = src[lp + "output.dense.weight"]
+        out[f"layers.{i}.b2"] = src[lp + "output.dense.bias"]
+    return out
+
+
+@torch.no_grad()
+def load_weights(model, path: str | Path) -> None:
+    src = _load_safetensors_dir(Path(path))
+    own = dict(model.named_parameters())
+    # direct native checkpoint?
+    if set(src).issuperset(own):
+        mapped = {k: src[k] for k in own}
+    else:
+        arch = model.dims.arch
+        if arch == "bert":
+            mapped = _map_bert(src, model)
+        elif arch in ("llama", "mistral"):
+            mapped = _map_llama(src, model)
+        elif arch == "esm2":
+            mapped = _map_esm2(src, model)
+        else:
+            raise ValueError(
+                f"no HF mapping for arch {arch}; provide a native checkpoint"
+            )
+    for name, p in own.items():
+        if name in mapped:
+            p.copy_(mapped[name].to(p.dtype))
+        else:
+            print(f"[loader] warning: no weight for {name}, keeping init")

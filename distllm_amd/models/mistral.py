@@ -1,0 +1,191 @@
+"""Mistral/Llama-class decoder on the distllm-amd HIP ops.
+
+One implementation serves both reference roles:
+  - SFR-Embedding-Mistral-style *embedder* (last_hidden_state +
+    last-token pooling; reference encoders/auto.py used with
+    examples/embed/leonardo-scaling/sfr-mistral.*.yaml), and
+  - Llama/Mistral *causal LM* for the generate path (reference delegated
+    to vLLM, generate/generators/vllm_backend.py; here: HIP paged-KV
+    decode + fused sampling).
+
+Pre-RMSNorm, RoPE (half-split), GQA, SwiGLU; bf16 weights, fp32 accum.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+from torch import nn
+
+from .. import ops
+from ..ops import torch_ref
+from .config import ModelDims
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, d: ModelDims):
+        super().__init__()
+        H = d.hidden_size
+        self.d = d
+        kvh = d.kv_heads
+        hd = d.head_dim
+        self.wq = nn.Parameter(torch.empty(H, H))
+        self.wk = nn.Parameter(torch.empty(kvh * hd, H))
+        self.wv = nn.Parameter(torch.empty(kvh * hd, H))
+        self.wo = nn.Parameter(torch.empty(H, H))
+        self.in_norm = nn.Parameter(torch.empty(H))
+        self.post_norm = nn.Parameter(torch.empty(H))
+        self.wgate = nn.Parameter(torch.empty(d.intermediate_size, H))
+        self.wup = nn.Parameter(torch.empty(d.intermediate_size, H))
+        self.wdown = nn.Parameter(torch.empty(H, d.intermediate_size))
+
+
+class DecoderModel(nn.Module):
+    """forward() -> last_hidden_state; generation via prefill()/decode()."""
+
+    def __init__(self, dims: ModelDims, dtype=torch.bfloat16,
+                 with_lm_head: bool | None = None):
+        super().__init__()
+        self.dims = dims
+        H = dims.hidden_size
+        self.tok_emb = nn.Parameter(torch.empty(dims.vocab_size, H))
+        self.layers = nn.ModuleList(
+            DecoderLayer(dims) for _ in range(dims.num_layers)
+        )
+        self.final_norm = nn.Parameter(torch.empty(H))
+        if with_lm_head is None:
+            with_lm_head = dims.arch == "llama"
+        self.lm_head = (
+            nn.Parameter(torch.empty(dims.vocab_size, H))
+            if with_lm_head and not dims.tie_word_embeddings
+            else None
+        )
+        self.with_lm_head = with_lm_head
+        self.reset_parameters()
+        self.to(dtype)
+        self._rope_cache: tuple[torch.Tensor, torch.Tensor] | None = None
+
+    @torch.no_grad()
+    def reset_parameters(self):
+        for n, p in self.named_parameters():
+            if p.dim() >= 2:
+                nn.init.normal_(p, std=0.02)
+            else:
+                nn.init.ones_(p)
+
+    @property
+    def device(self):
+        return self.tok_emb.device
+
+    @property
+    def dtype(self):
+        return self.tok_emb.dtype
+
+    @property
+    def embedding_size(self) -> int:
+        return self.dims.hidden_size
+
+    def _rope(self, device):
+        if self._rope_cache is None or self._rope_cache[0].device != device:
+            cs, sn = torch_ref.rope_tables(
+                self.dims.max_position_embeddings, self.dims.head_dim,
+                self.dims.rope_theta, device=device,
+            )
+            self._rope_cache = (cs, sn)
+        return self._rope_cache
+
+    def _attn_block(self, layer: DecoderLayer, x, mask_f, pos, kv_out=None):
+        d = self.dims
+        B, S, H = x.shape
+        nh, kvh, hd = d.num_heads, d.kv_heads, d.head_dim
+        cs, sn = self._rope(x.device)
+        h = ops.rms_norm(x, layer.in_norm, eps=d.rms_norm_eps)
+        q = ops.linear(h, layer.wq).view(B, S, nh, hd)
+        k = ops.linear(h, layer.wk).view(B, S, kvh, hd)
+        v = ops.linear(h, layer.wv).view(B, S, kvh, hd)
+        q = ops.rope_apply(q, cs, sn, pos)
+        k = ops.rope_apply(k, cs, sn, pos)
+        if kv_out is not None:
+            kv_out.append((k, v))
+        ctx = ops.attn_fwd(q, k, v, mask=mask_f, causal=True,
+                           scale=1.0 / math.sqrt(hd))
+        attn = ops.linear(ctx.reshape(B, S, H), layer.wo)
+        x = x + attn
+        h = ops.rms_norm(x, layer.post_norm, eps=d.rms_norm_eps)
+        gate = ops.linear(h, layer.wgate)
+        up = ops.linear(h, layer.wup)
+        x = x + ops.linear(ops.swiglu(gate, up), layer.wdown)
+        return x
+
+    @torch.no_grad()
+    def forward(self, input_ids: torch.Tensor, attention_mask: torch.Tensor,
+                kv_out: list | None = None) -> torch.Tensor:
+        ids = input_ids.long()
+        B, S = ids.shape
+        x = self.tok_emb[ids]
+        mask_f = torch.where(
+            attention_mask.bool(),
+            torch.zeros((), device=ids.device, dtype=torch.float32),
+            torch.full((), float("-inf"), device=ids.device,
+                       dtype=torch.float32),
+        )
+        pos = (
+            torch.arange(S, device=ids.device, dtype=torch.int32)
+            .repeat(B, 1)
+            .reshape(-1)
+        )
+        for layer in self.layers:
+            x = self._attn_block(layer, x, mask_f, pos, kv_out)
+        return ops.rms_norm(x, self.final_norm, eps=self.dims.rms_norm_eps)
+
+    @torch.no_grad()
+    def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        w = self.lm_head if self.lm_head is not None else self.tok_emb
+        return ops.linear(hidden, w).float()
+
+    # ---- generation-engine entry points --------------------------------
+    @torch.no_grad()
+    def prefill(self, input_ids, attention_mask):
+        """Full prefill; returns (last_hidden [B,S,H], kv list per layer
+        of (k, v) [B,S,HKV,D])."""
+        kv: list = []
+        hidden = self.forward(input_ids, attention_mask, kv_out=kv)
+        return hidden, kv
+
+    @torch.no_grad()
+    def decode_step(self, tokens, kcaches, vcaches, block_tables, ctx_lens):
+        """One decode step for B sequences.
+
+        tokens: [B] int; caches: per-layer paged [nblk, HKV, BLK, D];
+        ctx_lens: [B] int32 length INCLUDING the new token (the new
+        token's kv is written at position ctx_lens-1 by this call).
+        Returns logits [B, V] f32.
+        """
+        d = self.dims
+        B = tokens.shape[0]
+        nh, kvh, hd = d.num_heads, d.kv_heads, d.head_dim
+        cs, sn = self._rope(tokens.device)
+        x = self.tok_emb[tokens.long()].unsqueeze(1)  # [B,1,H]
+        pos = (ctx_lens - 1).int()
+        slot = pos
+        for li, layer in enumerate(self.layers):
+            h = ops.rms_norm(x, layer.in_norm, eps=d.rms_norm_eps)
+            q = ops.linear(h, layer.wq).view(B, 1, nh, hd)
+            k = ops.linear(h, layer.wk).view(B, 1, kvh, hd)
+            v = ops.linear(h, layer.wv).view(B, 1, kvh, hd)
+            q = ops.rope_apply(q, cs, sn, pos)
+            k = ops.rope_apply(k, cs, sn, pos)
+            ops.cache_put(k[:, 0], v[:, 0], kcaches[li], vcaches[li],
+                          block_tables, slot)
+            ctx = ops.paged_attn(q[:, 0].contiguous(), kcaches[li],
+                                 vcaches[li], block_tables, ctx_lens,
+                                 scale=1.0 / math.sqrt(hd))
+            attn = ops.linear(ctx.reshape(B, 1, d.hidden_size), layer.wo)
+            x = x + attn
+            h = ops.rms_norm(x, layer.post_norm, eps=d.rms_norm_eps)
+            gate = ops.linear(h, layer.wgate)
+            up = ops.linear(h, layer.wup)
+            x = x + ops.linear(ops.swiglu(gate, up), layer.wdown)
+        x = ops.rms_norm(x, self.final_norm, eps=d.rms_norm_eps)
+        return self.logits(x[:, 0])

@@ -164,8 +164,30 @@ def attn_fwd(Q, K, V, mask=None, causal=False, scale=None):
         scale = 1.0 / math.sqrt(D)
     if Q.is_cuda:
         _require_hip()
-        return _hip_ops.attn_fwd(Q.contiguous(), K.contiguous(),
-                                 V.contiguous(),
+        if D not in (64, 128):
+            # head_dim not natively tiled (e.g. small ESM2 variants,
+            # D=16/24/32/96): zero-pad to the next native width — exact,
+            # since padded dims add 0 to QK^T and produce 0 in O.
+            Dp = 64 if D < 64 else 128
+            pad = Dp - D
+            Qp = torch.nn.functional.pad(Q, (0, pad))
+            Kp = torch.nn.functional.pad(K, (0, pad))
+            Vp = torch.nn.functional.pad(V, (0, pad))
+            O = _hip_ops.attn_fwd(
+                Qp, Kp, Vp,
+                mask.contiguous() if mask is not None else None,
+                causal, scale,
+            )
+            return O[..., :D].contiguous()
+        if Q.stride(-1) != 1 or Q.stride(-2) != D:
+            Q = Q.contiguous()
+        if K.stride(-1) != 1 or K.stride(-2) != D:
+            K = K.contiguous()
+        if (V.stride(-1) != 1 or V.stride(-2) != D
+                or V.stride(0) != K.stride(0) or V.stride(1) != K.stride(1)):
+            V = V.contiguous()
+            K = K.contiguous()
+        return _hip_ops.attn_fwd(Q, K, V,
                                  mask.contiguous() if mask is not None else None,
                                  causal, scale)
     return torch_ref.attn_fwd(Q, K, V, mask, causal, scale)

@@ -47,7 +47,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
                 const unsigned short* __restrict__ Vt,
                 const float* __restrict__ maskp,
                 unsigned short* __restrict__ O, int Bsz, int SQ, int SKV,
-                int H, int HKV, float scale) {
+                int H, int HKV, float scale, long bstr_q, long tstr_q,
+                long bstr_k, long tstr_k) {
   constexpr int NCH = D / 8;        // 16B chunks per row
   constexpr int DBLK = D / 32;      // 32-wide d blocks
   constexpr int VROW = 40;          // padded V^T row, bf16 elems (80 B)
@@ -68,12 +69,13 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int b = bh / H;
   const int kvhead = head / (H / HKV);
 
-  const long qstride = (long)H * D;        // per-token stride in Q/O
-  const long kstride = (long)HKV * D;      // per-token stride in K/V
-  const unsigned short* Qb = Q + ((long)b * SQ) * qstride + (long)head * D;
-  const unsigned short* Kb = Kt + ((long)b * SKV) * kstride + (long)kvhead * D;
-  const unsigned short* Vb = Vt + ((long)b * SKV) * kstride + (long)kvhead * D;
-  unsigned short* Ob = O + ((long)b * SQ) * qstride + (long)head * D;
+  const long qstride = tstr_q;             // per-token stride in Q (elems)
+  const long kstride = tstr_k;             // per-token stride in K/V
+  const long ostride = (long)H * D;        // O is always contiguous BSHD
+  const unsigned short* Qb = Q + (long)b * bstr_q + (long)head * D;
+  const unsigned short* Kb = Kt + (long)b * bstr_k + (long)kvhead * D;
+  const unsigned short* Vb = Vt + (long)b * bstr_k + (long)kvhead * D;
+  unsigned short* Ob = O + ((long)b * SQ) * ostride + (long)head * D;
 
   const int q0w = qb * QBLK + wid * QWAVE;     // wave's first q row
   const int qrow_abs = q0w + qcol;             // this lane's q row
@@ -298,7 +300,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
         float v1 = oimg[q * (D + 1) + lane * 2 + 1];
         unsigned int pk = ((unsigned int)f2bf(v0)) |
                           ((unsigned int)f2bf(v1) << 16);
-        *reinterpret_cast<unsigned int*>(Ob + (long)qa * qstride + lane * 2) =
+        *reinterpret_cast<unsigned int*>(Ob + (long)qa * ostride + lane * 2) =
             pk;
       }
     } else {
@@ -311,7 +313,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
           float v1 = oimg[(q + h) * (D + 1) + col + 1];
           unsigned int pk = ((unsigned int)f2bf(v0)) |
                             ((unsigned int)f2bf(v1) << 16);
-          *reinterpret_cast<unsigned int*>(Ob + (long)qa * qstride + col) = pk;
+          *reinterpret_cast<unsigned int*>(Ob + (long)qa * ostride + col) = pk;
         }
       }
     }
@@ -321,7 +323,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
 extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
                                 const float* mask, void* O, int B, int SQ,
                                 int SKV, int H, int HKV, int D, bool causal,
-                                float scale, hipStream_t stream) {
+                                float scale, long bstr_q, long tstr_q,
+                                long bstr_k, long tstr_k, hipStream_t stream) {
   const int nqb = (SQ + QBLK - 1) / QBLK;
   const int grid = B * H * nqb;
   // LDS: staging (2K + 2V + 2mask) or epilogue (4 waves * 32 * (D+1) f32)
@@ -334,7 +337,8 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
   hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM>), dim3(grid), dim3(256),     \
                      lds, stream, (const unsigned short*)Q,                    \
                      (const unsigned short*)K, (const unsigned short*)V, mask, \
-                     (unsigned short*)O, B, SQ, SKV, H, HKV, scale)
+                     (unsigned short*)O, B, SQ, SKV, H, HKV, scale, bstr_q,    \
+                     tstr_q, bstr_k, tstr_k)
   const bool hm = mask != nullptr;
   if (D == 64) {
     if (causal && hm) ATTN_DISPATCH(64, true, true);

@@ -34,8 +34,8 @@ void launch_adjacent_cosine(const float*, float*, int, int, hipStream_t);
 void launch_gemm_bf16(const void*, const void*, const void*, void*, int, int,
                       int, int, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, const float*,
-                     void*, int, int, int, int, int, int, bool, float,
-                     hipStream_t);
+                     void*, int, int, int, int, int, int, bool, float, long,
+                     long, long, long, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, const int*,
                  long, int, int, hipStream_t);
 void launch_topk_chunk(const float*, const int*, float*, int*, int, long, int,
@@ -158,13 +158,21 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor B,
 torch::Tensor attn_fwd(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
                        c10::optional<torch::Tensor> mask, bool causal,
                        double scale) {
-  CHECK_DEV(Q); CHECK_CONT(Q); CHECK_BF16(Q);
-  CHECK_DEV(K); CHECK_CONT(K); CHECK_BF16(K);
-  CHECK_DEV(V); CHECK_CONT(V); CHECK_BF16(V);
+  // Q/K/V may be strided views (e.g. slices of a packed QKV tensor) as
+  // long as the innermost (D) is contiguous and heads are D-strided.
+  CHECK_DEV(Q); CHECK_BF16(Q);
+  CHECK_DEV(K); CHECK_BF16(K);
+  CHECK_DEV(V); CHECK_BF16(V);
   const int B = Q.size(0), SQ = Q.size(1), H = Q.size(2), D = Q.size(3);
   const int SKV = K.size(1), HKV = K.size(2);
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   TORCH_CHECK(H % HKV == 0, "GQA requires H % HKV == 0");
+  TORCH_CHECK(Q.stride(3) == 1 && K.stride(3) == 1 && V.stride(3) == 1,
+              "innermost dim must be contiguous");
+  TORCH_CHECK(Q.stride(2) == D && K.stride(2) == D && V.stride(2) == D,
+              "head stride must be D");
+  TORCH_CHECK(K.stride(0) == V.stride(0) && K.stride(1) == V.stride(1),
+              "K/V must share layout");
   const float* mp = nullptr;
   if (mask) {
     CHECK_DEV((*mask)); CHECK_CONT((*mask)); CHECK_F32((*mask));
@@ -172,9 +180,11 @@ torch::Tensor attn_fwd(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
                 "mask must be [B, SKV] f32");
     mp = mask->data_ptr<float>();
   }
-  auto O = torch::empty_like(Q);
+  auto O = torch::empty({B, SQ, H, D}, Q.options());
   launch_attn_fwd(Q.data_ptr(), K.data_ptr(), V.data_ptr(), mp, O.data_ptr(),
-                  B, SQ, SKV, H, HKV, D, causal, (float)scale, cur_stream());
+                  B, SQ, SKV, H, HKV, D, causal, (float)scale,
+                  (long)Q.stride(0), (long)Q.stride(1), (long)K.stride(0),
+                  (long)K.stride(1), cur_stream());
   return O;
 }
 
