@@ -48,6 +48,15 @@ class EsmTokenizer:
         self.padding_side = "right"
         self.vocab_size = len(ESM_VOCAB)
 
+    def decode(self, ids, skip_special_tokens=True):
+        toks = []
+        for i in ids:
+            t = ESM_VOCAB[int(i)] if 0 <= int(i) < len(ESM_VOCAB) else "<unk>"
+            if skip_special_tokens and t.startswith("<"):
+                continue
+            toks.append(t)
+        return "".join(toks)
+
     def __call__(self, texts, padding=True, truncation=True, max_length=None,
                  return_tensors="pt", pad_to_multiple_of=None, **kw):
         if isinstance(texts, str):
@@ -81,6 +90,19 @@ class HashWordTokenizer:
     def _wid(self, w: str) -> int:
         h = hashlib.md5(w.lower().encode()).digest()
         return 4 + int.from_bytes(h[:4], "little") % (self.vocab_size - 4)
+
+    @property
+    def eos_token_id(self):
+        return self.sep_token_id
+
+    def decode(self, ids, skip_special_tokens=True):
+        out = []
+        for i in ids:
+            i = int(i)
+            if skip_special_tokens and i < 4:
+                continue
+            out.append(f"tok{i}")
+        return " ".join(out)
 
     def __call__(self, texts, padding=True, truncation=True, max_length=None,
                  return_tensors="pt", pad_to_multiple_of=None, **kw):
