@@ -1,0 +1,181 @@
+"""Generator strategies: the native MI355X engine (replacing vLLM) and
+an OpenAI-compatible HTTP client (replacing the reference's
+HTTP-vLLM/argo/openai backends).
+
+Reference parity: distllm/generate/generators/* — the `vllm` config
+name is accepted as an alias of `native` so reference YAMLs keep
+working; sampling defaults match vllm_backend.py:46-60 (temperature .5,
+max_tokens 2000, min_p .1).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Any, Literal, Union
+
+from pydantic import Field
+
+from ..registry import registry
+from ..utils import BaseConfig, retry
+
+
+class NativeGeneratorConfig(BaseConfig):
+    """In-process generation on the native HIP engine."""
+
+    name: Literal["native", "vllm", "huggingface"] = "native"
+    model: str = "llama-tiny"
+    weights_path: str = ""
+    tokenizer_path: str = ""
+    dtype: str = "bfloat16"
+    temperature: float = 0.5
+    min_p: float = 0.1
+    top_p: float = 0.0
+    max_tokens: int = 2000
+    max_batch_size: int = 64
+    max_model_len: int = 4096
+    # accepted for reference-YAML compatibility; single-GPU engine per
+    # rank (data-parallel across ranks) in this round:
+    tensor_parallel_size: int = 1
+    trust_remote_code: bool = True
+    quantization: bool = False
+
+
+class NativeGenerator:
+    def __init__(self, config: NativeGeneratorConfig):
+        import torch
+
+        from ..embed.encoders import resolve_preset
+        from ..embed.tokenizers import load_tokenizer
+        from ..engine import LLMEngine, SamplingParams
+        from ..models import build_model, get_dims
+
+        self.config = config
+        preset = resolve_preset(config.model)
+        dims = get_dims(preset)
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+        dtype = torch.bfloat16 if device != "cpu" else torch.float32
+        self.model = build_model(preset, device=device, dtype=dtype)
+        if config.weights_path:
+            from ..models.loader import load_weights
+
+            load_weights(self.model, config.weights_path)
+        self.tokenizer = load_tokenizer(
+            config.tokenizer_path or config.model,
+            arch=dims.arch, vocab_size=dims.vocab_size,
+            model_max_length=dims.max_position_embeddings,
+        )
+        self.engine = LLMEngine(self.model,
+                                max_batch_size=config.max_batch_size,
+                                max_model_len=config.max_model_len)
+        eos = getattr(self.tokenizer, "eos_token_id", None)
+        self.params = SamplingParams(
+            temperature=config.temperature,
+            min_p=config.min_p,
+            top_p=config.top_p if config.top_p > 0 else 1.0,
+            max_tokens=config.max_tokens,
+            stop_token_ids=(eos,) if eos is not None else (),
+        )
+
+    def generate(self, prompts: list[str]) -> list[str]:
+        if isinstance(prompts, str):
+            prompts = [prompts]
+        enc = self.tokenizer(prompts, padding=False, truncation=True,
+                             return_tensors=None)
+        ids = enc["input_ids"]
+        # tokenizer may return tensors or lists depending on provider
+        import torch
+
+        if isinstance(ids, torch.Tensor):
+            mask = enc["attention_mask"]
+            prompt_ids = [
+                ids[i][mask[i].bool()].tolist() for i in range(len(prompts))
+            ]
+        else:
+            prompt_ids = [list(x) for x in ids]
+        out_ids = self.engine.generate(prompt_ids, self.params)
+        return [
+            self.tokenizer.decode(o, skip_special_tokens=True)
+            for o in out_ids
+        ]
+
+
+class OpenAICompatGeneratorConfig(BaseConfig):
+    """HTTP client against any OpenAI-compatible /v1 server (our
+    chat_server, a remote endpoint, or an Argo-style proxy)."""
+
+    name: Literal["openai_http", "argo", "langchain"] = "openai_http"
+    base_url: str = "http://localhost:8000/v1"
+    model: str = "default"
+    api_key_env: str = Field(
+        "OPENAI_API_KEY", description="env var holding the key, if any"
+    )
+    temperature: float = 0.5
+    max_tokens: int = 2000
+    timeout: float = 120.0
+    use_chat_endpoint: bool = True
+
+
+class OpenAICompatGenerator:
+    def __init__(self, config: OpenAICompatGeneratorConfig):
+        import httpx
+
+        self.config = config
+        headers = {}
+        key = os.environ.get(config.api_key_env, "")
+        if key:
+            headers["Authorization"] = f"Bearer {key}"
+        self._client = httpx.Client(base_url=config.base_url,
+                                    headers=headers,
+                                    timeout=config.timeout)
+
+    @retry(tries=5, base_delay=1.0)
+    def _one(self, prompt: str) -> str:
+        c = self.config
+        if c.use_chat_endpoint:
+            r = self._client.post("/chat/completions", json={
+                "model": c.model,
+                "messages": [{"role": "user", "content": prompt}],
+                "temperature": c.temperature,
+                "max_tokens": c.max_tokens,
+            })
+            r.raise_for_status()
+            return r.json()["choices"][0]["message"]["content"]
+        r = self._client.post("/completions", json={
+            "model": c.model, "prompt": prompt,
+            "temperature": c.temperature, "max_tokens": c.max_tokens,
+        })
+        r.raise_for_status()
+        return r.json()["choices"][0]["text"]
+
+    def generate(self, prompts: list[str]) -> list[str]:
+        if isinstance(prompts, str):
+            prompts = [prompts]
+        return [self._one(p) for p in prompts]
+
+
+GeneratorConfigs = Union[NativeGeneratorConfig, OpenAICompatGeneratorConfig]
+
+STRATEGIES: dict[str, tuple[type, type]] = {
+    "native": (NativeGeneratorConfig, NativeGenerator),
+    "vllm": (NativeGeneratorConfig, NativeGenerator),
+    "huggingface": (NativeGeneratorConfig, NativeGenerator),
+    "openai_http": (OpenAICompatGeneratorConfig, OpenAICompatGenerator),
+    "argo": (OpenAICompatGeneratorConfig, OpenAICompatGenerator),
+    "langchain": (OpenAICompatGeneratorConfig, OpenAICompatGenerator),
+}
+
+
+def _factory(kwargs: dict[str, Any]):
+    name = kwargs.get("name", "")
+    if name not in STRATEGIES:
+        raise ValueError(
+            f"unknown generator name {name!r}; options: {sorted(STRATEGIES)}"
+        )
+    cfg_cls, impl_cls = STRATEGIES[name]
+    return impl_cls(cfg_cls(**kwargs))
+
+
+def get_generator(kwargs: dict[str, Any], register: bool = False):
+    if register:
+        return registry.get(_factory, dict(kwargs))
+    return _factory(dict(kwargs))
