@@ -1,0 +1,205 @@
+"""Typer CLI: the user-facing verbs of the framework.
+
+Reference parity: distllm/cli.py (embed, merge, generate, tokenize,
+chunk-fasta-file) with the same flag names; plus chat / serve / eval
+entry points for the L6 applications.
+Entry point: `distllm = distllm_amd.cli:main` (pyproject) or
+`python -m distllm_amd.cli`.
+"""
+
+from __future__ import annotations
+
+from pathlib import Path
+
+import typer
+
+app = typer.Typer(no_args_is_help=True, pretty_exceptions_enable=False)
+
+
+@app.command()
+def embed(
+    encoder_name: str = typer.Option("auto", help="auto|esm2|esmc"),
+    pretrained_model_name_or_path: str = typer.Option(
+        "pubmedbert-base", help="model preset or local checkpoint"
+    ),
+    data_path: Path = typer.Option(..., help="input file or directory"),
+    data_extension: str = typer.Option("jsonl"),
+    output_path: Path = typer.Option(...),
+    dataset_name: str = typer.Option(
+        "jsonl_chunk",
+        help="jsonl|jsonl_chunk|fasta|sequence_per_line|huggingface",
+    ),
+    batch_size: int = typer.Option(8),
+    chunk_batch_size: int = typer.Option(8),
+    buffer_size: int = typer.Option(1),
+    pooler_name: str = typer.Option("mean", help="mean|last_token"),
+    embedder_name: str = typer.Option(
+        "semantic_chunk", help="full_sequence|semantic_chunk"
+    ),
+    writer_name: str = typer.Option("huggingface", help="huggingface|numpy"),
+    half_precision: bool = typer.Option(True),
+    eval_mode: bool = typer.Option(True),
+    compile_model: bool = typer.Option(False),
+    quantization: bool = typer.Option(False),
+) -> None:
+    """Embed a file/directory serially in-process (single GPU)."""
+    from .distributed_embedding import embedding_worker
+
+    dataset_kwargs: dict = {"name": dataset_name, "batch_size": batch_size}
+    if dataset_name == "jsonl_chunk":
+        dataset_kwargs["buffer_size"] = buffer_size
+    encoder_kwargs = {
+        "name": encoder_name,
+        "pretrained_model_name_or_path": pretrained_model_name_or_path,
+        "half_precision": half_precision,
+    }
+    embedder_kwargs: dict = {"name": embedder_name}
+    if embedder_name == "semantic_chunk":
+        embedder_kwargs["chunk_batch_size"] = chunk_batch_size
+
+    files = (
+        [data_path]
+        if data_path.is_file()
+        else sorted(data_path.glob(f"*.{data_extension}"))
+    )
+    output_path.mkdir(exist_ok=True, parents=True)
+    for f in files:
+        embedding_worker(
+            f, output_path, dataset_kwargs, encoder_kwargs,
+            {"name": pooler_name}, embedder_kwargs, {"name": writer_name},
+        )
+
+
+@app.command()
+def merge(
+    writer_name: str = typer.Option("huggingface"),
+    num_proc: int = typer.Option(1),
+    dataset_dir: Path = typer.Option(..., help="dir of dataset shards"),
+    output_dir: Path = typer.Option(...),
+) -> None:
+    """Merge per-file dataset shards into one dataset."""
+    from .embed import get_writer
+
+    writer = get_writer({"name": writer_name})
+    shards = sorted(p for p in Path(dataset_dir).iterdir() if p.is_dir())
+    writer.merge(shards, output_dir, num_proc=num_proc)
+    typer.echo(f"merged {len(shards)} shards -> {output_dir}")
+
+
+@app.command()
+def generate(
+    input_dir: Path = typer.Option(...),
+    output_dir: Path = typer.Option(...),
+    prompt_name: str = typer.Option("question_chunk"),
+    reader_name: str = typer.Option("huggingface"),
+    writer_name: str = typer.Option("huggingface"),
+    generator_name: str = typer.Option("native", help="native|vllm|openai_http"),
+    llm_name: str = typer.Option("llama3-8b"),
+    temperature: float = typer.Option(0.5),
+    min_p: float = typer.Option(0.1),
+    top_p: float = typer.Option(0.0),
+    max_tokens: int = typer.Option(2000),
+    batch_size: int = typer.Option(64),
+    quantization: bool = typer.Option(False),
+) -> None:
+    """Generate over a directory of inputs serially in-process."""
+    from .distributed_generation import generate_worker
+
+    generator_kwargs = {
+        "name": generator_name,
+        "model": llm_name,
+        "temperature": temperature,
+        "min_p": min_p,
+        "top_p": top_p,
+        "max_tokens": max_tokens,
+        "max_batch_size": batch_size,
+    }
+    if generator_name in ("openai_http", "argo", "langchain"):
+        generator_kwargs = {"name": generator_name, "model": llm_name,
+                            "temperature": temperature,
+                            "max_tokens": max_tokens}
+    output_dir.mkdir(parents=True, exist_ok=True)
+    for f in sorted(Path(input_dir).iterdir()):
+        generate_worker(f, output_dir, {"name": prompt_name},
+                        {"name": reader_name}, {"name": writer_name},
+                        generator_kwargs)
+
+
+@app.command()
+def tokenize(
+    input_dir: Path = typer.Option(...),
+    output_dir: Path = typer.Option(...),
+    text_field: str = typer.Option("text"),
+    tokenizer_name: str = typer.Option("pubmedbert-base"),
+    save_labels: bool = typer.Option(False),
+) -> None:
+    """Tokenize jsonl files into HF datasets."""
+    from .distributed_tokenization import TokenizationConfig, main as tmain
+
+    tmain(TokenizationConfig(
+        input_dir=input_dir, output_dir=output_dir, text_field=text_field,
+        tokenizer_name=tokenizer_name, save_labels=save_labels,
+    ))
+
+
+@app.command()
+def chunk_fasta_file(
+    input_file: Path = typer.Option(...),
+    output_dir: Path = typer.Option(...),
+    num_chunks: int = typer.Option(8),
+) -> None:
+    """Split one fasta file into N chunk files."""
+    from .embed.datasets import read_fasta, write_fasta
+    from .utils import batch_data
+
+    seqs = read_fasta(input_file)
+    output_dir.mkdir(parents=True, exist_ok=True)
+    per = max(1, (len(seqs) + num_chunks - 1) // num_chunks)
+    for i, chunk in enumerate(batch_data(seqs, per)):
+        write_fasta(chunk, output_dir / f"{input_file.stem}_{i}.fasta")
+    typer.echo(f"wrote {min(num_chunks, len(seqs))} chunks")
+
+
+@app.command()
+def chat(
+    config: Path = typer.Option(..., help="chat YAML config"),
+) -> None:
+    """Terminal RAG chat."""
+    from .chat import chat_with_model, ChatAppConfig
+
+    chat_with_model(ChatAppConfig.from_yaml(config))
+
+
+@app.command()
+def serve(
+    config: Path = typer.Option(..., help="chat YAML config"),
+    host: str = typer.Option("0.0.0.0"),
+    port: int = typer.Option(8000),
+) -> None:
+    """OpenAI-compatible RAG chat server."""
+    import os
+
+    import uvicorn
+
+    os.environ["DISTLLM_CHAT_CONFIG"] = str(config)
+    from .chat_server import app as fastapi_app
+
+    uvicorn.run(fastapi_app, host=host, port=port)
+
+
+@app.command()
+def evaluate(
+    config: Path = typer.Option(..., help="eval-suite YAML config"),
+) -> None:
+    """Run the RAG eval suite."""
+    from .rag.evaluate import EvalSuiteConfig, run_eval_suite
+
+    run_eval_suite(EvalSuiteConfig.from_yaml(config))
+
+
+def main() -> None:
+    app()
+
+
+if __name__ == "__main__":
+    main()

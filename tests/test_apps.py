@@ -1,0 +1,174 @@
+"""CPU tests for L6 applications: chat, OpenAI-compatible server,
+MCQA harness components, CLI verbs."""
+
+import json
+
+import pytest
+import yaml
+from typer.testing import CliRunner
+
+from distllm_amd.mcqa.harness import (
+    Checkpointer,
+    compute_metrics,
+    find_free_port,
+    make_chunk_id,
+    parse_grade_json,
+    question_hash,
+)
+
+
+def test_parse_grade_json_ladder():
+    assert parse_grade_json('{"correct": true, "reason": "ok"}')["correct"]
+    assert parse_grade_json(
+        'Sure! Here: {"correct": false, "reason": "wrong"} hope that helps'
+    )["correct"] is False
+    assert parse_grade_json('I think "correct": true overall')["correct"]
+    assert parse_grade_json("no json at all") is None
+
+
+def test_chunk_id_and_hash_stability():
+    a = make_chunk_id("/data/p1.jsonl", 7)
+    assert a == make_chunk_id("/data/p1.jsonl", 7)
+    assert a != make_chunk_id("/data/p2.jsonl", 7)
+    assert len(a.split("_")[0]) == 16
+    assert question_hash("Q?") == question_hash("Q?")
+
+
+def test_checkpointer_roundtrip(tmp_path):
+    ck = Checkpointer(tmp_path, tmp_path / "qs.json", "modelA")
+    cfg = {"questions_file": str(tmp_path / "qs.json")}
+    res = {0: {"answer": "a"}, 2: {"answer": "b"}}
+    p = ck.save(res, cfg)
+    assert p.exists()
+    found = ck.find_latest()
+    assert found == p
+    loaded = ck.load(found, cfg)
+    assert loaded == res
+    # mismatched questions file -> ignored
+    assert ck.load(found, {"questions_file": "other.json"}) == {}
+
+
+def test_metrics():
+    rows = [
+        {"grade": {"correct": True}, "type": "mc",
+         "retrieval": {"used_rag": True, "source_chunk_retrieved": True,
+                       "question_hash_retrieved": False}, "elapsed_s": 1.0},
+        {"grade": {"correct": False}, "type": "qa",
+         "retrieval": {"used_rag": True, "source_chunk_retrieved": False,
+                       "question_hash_retrieved": True}, "elapsed_s": 2.0},
+    ]
+    m = compute_metrics(rows)
+    assert m["overall_accuracy"] == 0.5
+    assert m["mc_accuracy"] == 1.0
+    assert m["qa_accuracy"] == 0.0
+    assert m["source_retrieval_rate"] == 0.5
+    assert m["question_hash_retrieval_rate"] == 0.5
+
+
+def test_find_free_port():
+    p = find_free_port()
+    assert 1024 <= p <= 65535
+
+
+@pytest.fixture()
+def chat_config(tmp_path):
+    cfg = {
+        "generator_config": {
+            "name": "native", "model": "llama-tiny",
+            "temperature": 0.0, "min_p": 0.0, "max_tokens": 4,
+            "max_model_len": 256, "max_batch_size": 2,
+        },
+        "system_prompt": "Test assistant.",
+        "transcript_dir": str(tmp_path / "transcripts"),
+    }
+    p = tmp_path / "chat.yaml"
+    p.write_text(yaml.safe_dump(cfg))
+    return p
+
+
+def test_chat_session_and_transcript(chat_config, tmp_path):
+    from distllm_amd.chat import ChatAppConfig, ChatSession
+
+    session = ChatSession(ChatAppConfig.from_yaml(chat_config))
+    ans = session.turn("Hello there")
+    assert isinstance(ans, str)
+    assert session.template.history[-1][0] == "assistant"
+    out = session.save_transcript()
+    assert out is not None and out.exists()
+
+
+def test_chat_server_endpoint(chat_config, monkeypatch):
+    from fastapi.testclient import TestClient
+
+    import distllm_amd.chat_server as srv
+
+    monkeypatch.setenv("DISTLLM_CHAT_CONFIG", str(chat_config))
+    srv._session = None
+    client = TestClient(srv.app)
+    assert client.get("/health").json()["status"] == "ok"
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "Hi"}],
+    })
+    assert r.status_code == 200
+    body = r.json()
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    # streaming shim
+    r2 = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "Hi"}], "stream": True,
+    })
+    assert r2.status_code == 200
+    assert "data:" in r2.text and "[DONE]" in r2.text
+
+
+def test_env_substitution(monkeypatch):
+    from distllm_amd.chat import substitute_env
+
+    monkeypatch.setenv("MY_KEY", "sekrit")
+    obj = {"a": "${env:MY_KEY}", "b": ["${env:MY_KEY}", 3]}
+    out = substitute_env(obj)
+    assert out == {"a": "sekrit", "b": ["sekrit", 3]}
+
+
+def test_cli_embed_and_merge(tmp_path):
+    from distllm_amd.cli import app
+
+    runner = CliRunner()
+    data = tmp_path / "in"
+    data.mkdir()
+    with open(data / "a.jsonl", "w") as fh:
+        for i in range(3):
+            fh.write(json.dumps({"text": f"Sample text {i}.",
+                                 "path": f"p{i}"}) + "\n")
+    out = tmp_path / "out"
+    r = runner.invoke(app, [
+        "embed", "--data-path", str(data), "--output-path", str(out),
+        "--pretrained-model-name-or-path", "bert-tiny",
+        "--dataset-name", "jsonl", "--embedder-name", "full_sequence",
+        "--no-half-precision",
+    ])
+    assert r.exit_code == 0, r.output
+    shards = [p for p in out.iterdir() if p.is_dir()]
+    assert len(shards) == 1
+    merged = tmp_path / "merged"
+    r2 = runner.invoke(app, [
+        "merge", "--dataset-dir", str(out), "--output-dir", str(merged),
+    ])
+    assert r2.exit_code == 0, r2.output
+    from datasets import load_from_disk
+
+    assert len(load_from_disk(str(merged))) == 3
+
+
+def test_cli_chunk_fasta(tmp_path):
+    from distllm_amd.cli import app
+
+    runner = CliRunner()
+    f = tmp_path / "x.fasta"
+    f.write_text(">a\nMKV\n>b\nACD\n>c\nGLY\n>d\nWYQ\n")
+    out = tmp_path / "chunks"
+    r = runner.invoke(app, [
+        "chunk-fasta-file", "--input-file", str(f), "--output-dir",
+        str(out), "--num-chunks", "2",
+    ])
+    assert r.exit_code == 0, r.output
+    assert len(list(out.glob("*.fasta"))) == 2
