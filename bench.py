@@ -30,7 +30,16 @@ def main() -> None:
     ap.add_argument("--batch-size", type=int, default=512)
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--model", type=str, default="pubmedbert-base")
+    ap.add_argument("--mode", type=str, default="embed",
+                    choices=["embed", "generate"],
+                    help="embed = flagship seqs/s; generate = decode "
+                         "tokens/s (Llama-class, paged KV)")
+    ap.add_argument("--prompt-len", type=int, default=128)
     args = ap.parse_args()
+
+    if args.mode == "generate":
+        bench_generate(args)
+        return
 
     from distllm_amd.models import build_model
     from distllm_amd.parallel import (barrier, get_rank, get_world_size,
@@ -112,6 +121,93 @@ def main() -> None:
                 "model": args.model,
                 "global_batch": world * B,
                 "seq_len": S,
+                "parallelism": f"dp{world}",
+            },
+        }))
+
+
+def bench_generate(args) -> None:
+    """Decode throughput: one step = one continuous-batching decode step
+    for the whole batch (paged-KV attention + sampling)."""
+    import torch
+
+    from distllm_amd import ops
+    from distllm_amd.engine import LLMEngine, SamplingParams
+    from distllm_amd.models import build_model
+    from distllm_amd.parallel import barrier, init_distributed
+
+    rank, world = init_distributed()
+    have_gpu = torch.cuda.is_available()
+    if have_gpu:
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+        device, dtype = f"cuda:{local_rank}", torch.bfloat16
+    else:
+        device, dtype = "cpu", torch.float32
+
+    model_name = args.model if args.model != "pubmedbert-base" \
+        else "llama3-8b"
+    torch.manual_seed(99 + rank)
+    model = build_model(model_name, device=device, dtype=dtype)
+    B, P = args.batch_size, args.prompt_len
+    engine = LLMEngine(model, max_batch_size=B,
+                       max_model_len=P + args.steps + args.warmup + 8)
+    params = SamplingParams(temperature=0.5, min_p=0.1,
+                            max_tokens=args.steps + args.warmup + 4,
+                            stop_token_ids=())
+
+    # set up B running sequences via one prefill
+    prompts = torch.randint(4, model.dims.vocab_size, (B, P)).tolist()
+    from distllm_amd.engine.engine import _Seq
+
+    seqs = [_Seq(seq_id=i, prompt_ids=p) for i, p in enumerate(prompts)]
+    for s in seqs:
+        s.blocks = engine.blockman.alloc(
+            engine.blockman.blocks_for(P + params.max_tokens + 1))
+    results: dict = {}
+    engine._run_prefill(seqs, params, results)
+
+    def step():
+        engine._run_decode(seqs, params, results)
+
+    for _ in range(args.warmup):
+        step()
+    barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if have_gpu:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    value = world * B * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": f"generate tokens/sec ({model_name} decode)",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if have_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": world * B,
+                "seq_len": P,
                 "parallelism": f"dp{world}",
             },
         }))
