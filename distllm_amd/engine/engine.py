@@ -51,11 +51,12 @@ class _Seq:
 
 
 class BlockManager:
-    """Fixed-pool paged KV block allocator (BLK tokens per block)."""
+    """Fixed-pool paged KV block allocator (BLK tokens per block).
+    Block 0 is reserved as the dummy target for padded graph slots."""
 
     def __init__(self, num_blocks: int, blk: int = 16):
         self.blk = blk
-        self.free: list[int] = list(range(num_blocks))
+        self.free: list[int] = list(range(1, num_blocks))
 
     def can_alloc(self, n: int) -> bool:
         return len(self.free) >= n
@@ -80,7 +81,8 @@ class LLMEngine:
     def __init__(self, model: DecoderModel, max_batch_size: int = 64,
                  max_model_len: int = 4096,
                  kv_memory_fraction: float = 0.6,
-                 num_blocks: int | None = None):
+                 num_blocks: int | None = None,
+                 use_graphs: bool | None = None):
         self.model = model
         d = model.dims
         self.device = model.device
@@ -108,6 +110,13 @@ class LLMEngine:
         self.vcaches = [torch.zeros_like(k) for k in self.kcaches]
         self.blockman = BlockManager(num_blocks, self.BLK)
         self._next_id = 0
+        # hipGraph-captured decode (one graph per pow2 batch bucket):
+        # the decode step is ~300 small launches per step for a 32-layer
+        # model — graph replay removes the launch overhead.
+        self.use_graphs = (use_graphs if use_graphs is not None
+                           else str(dev).startswith("cuda"))
+        self.max_blocks_per_seq = (self.max_model_len // self.BLK) + 1
+        self._graphs: dict[int, "_GraphRunner"] = {}
 
     # ---- public --------------------------------------------------------
     @torch.no_grad()
@@ -154,8 +163,9 @@ class LLMEngine:
 
         return [results.get(i, []) for i in range(len(prompts_ids))]
 
-    def _block_table(self, seqs: list[_Seq]) -> torch.Tensor:
-        maxb = max(len(s.blocks) for s in seqs)
+    def _block_table(self, seqs: list[_Seq],
+                     width: int | None = None) -> torch.Tensor:
+        maxb = width or max(len(s.blocks) for s in seqs)
         bt = torch.zeros(len(seqs), maxb, dtype=torch.int32,
                          device=self.device)
         for i, s in enumerate(seqs):
@@ -200,15 +210,29 @@ class LLMEngine:
         self._sample_and_append(seqs, logits, params, results)
 
     @torch.no_grad()
+    def _graph_for(self, n: int) -> "_GraphRunner":
+        bucket = 1
+        while bucket < n:
+            bucket <<= 1
+        bucket = min(max(bucket, n), max(self.max_batch_size, n))
+        if bucket not in self._graphs:
+            self._graphs[bucket] = _GraphRunner(self, bucket)
+        return self._graphs[bucket]
+
     def _run_decode(self, seqs: list[_Seq], params: SamplingParams,
                     results: dict[int, list[int]]) -> None:
         dev = self.device
         tokens = torch.tensor([s.out_ids[-1] for s in seqs], device=dev)
-        bt = self._block_table(seqs)
         ctx = torch.tensor([s.length for s in seqs], dtype=torch.int32,
                            device=dev)
-        logits = self.model.decode_step(tokens, self.kcaches, self.vcaches,
-                                        bt, ctx)
+        if self.use_graphs:
+            bt = self._block_table(seqs, width=self.max_blocks_per_seq)
+            logits = self._graph_for(len(seqs)).run(tokens, bt, ctx)
+            logits = logits[: len(seqs)]
+        else:
+            bt = self._block_table(seqs)
+            logits = self.model.decode_step(tokens, self.kcaches,
+                                            self.vcaches, bt, ctx)
         self._sample_and_append(seqs, logits, params, results)
 
     def _sample_and_append(self, seqs, logits, params, results):
@@ -236,6 +260,13 @@ class LLMEngine:
                 self.blockman.release(s.blocks)
                 s.blocks = []
 
+    def warm_graphs(self, batch_sizes: list[int] | None = None) -> None:
+        """Pre-capture decode graphs for the given batch buckets."""
+        if not self.use_graphs:
+            return
+        for b in batch_sizes or [self.max_batch_size]:
+            self._graph_for(b)
+
     def _sample_top_p(self, logits: torch.Tensor,
                       params: SamplingParams) -> torch.Tensor:
         t = max(params.temperature, 1e-5)
@@ -247,3 +278,52 @@ class LLMEngine:
         sorted_p = sorted_p / sorted_p.sum(-1, keepdim=True)
         choice = torch.multinomial(sorted_p, 1)
         return sorted_i.gather(1, choice).squeeze(1).int()
+
+
+class _GraphRunner:
+    """One hipGraph-captured decode step at a fixed batch size.
+
+    Static input/output tensors; padded slots point at the reserved
+    dummy KV block 0 with ctx_len=1, so their cache writes and reads are
+    harmless. Capture follows the torch.cuda.graph protocol (side-stream
+    warmup, then capture); replay removes the ~10 launches/layer CPU
+    overhead that dominates small-batch decode.
+    """
+
+    def __init__(self, engine: LLMEngine, batch: int):
+        dev = engine.device
+        self.engine = engine
+        self.batch = batch
+        self.tokens = torch.zeros(batch, dtype=torch.long, device=dev)
+        self.bt = torch.zeros(batch, engine.max_blocks_per_seq,
+                              dtype=torch.int32, device=dev)
+        self.ctx = torch.ones(batch, dtype=torch.int32, device=dev)
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warmup allocations/paths
+                engine.model.decode_step(self.tokens, engine.kcaches,
+                                         engine.vcaches, self.bt, self.ctx)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.logits = engine.model.decode_step(
+                self.tokens, engine.kcaches, engine.vcaches, self.bt,
+                self.ctx)
+
+    def run(self, tokens: torch.Tensor, bt: torch.Tensor,
+            ctx: torch.Tensor) -> torch.Tensor:
+        n = tokens.shape[0]
+        self.tokens[:n] = tokens
+        if n < self.batch:
+            self.tokens[n:] = 0
+        self.bt[:n] = bt
+        if n < self.batch:
+            self.bt[n:] = 0
+        self.ctx[:n] = ctx
+        if n < self.batch:
+            self.ctx[n:] = 1
+        self.graph.replay()
+        return self.logits
