@@ -5,6 +5,9 @@ shape and prints effective TFLOP/s. Used with rocprofv3 for PMC runs.
 """
 
 import argparse
+import sys
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).parent.parent))
 import time
 
 import torch
