@@ -1,0 +1,119 @@
+"""Foundation + misc parity tests: Timer log round-trip, registry
+warm-start, tokenization entry, MCQA config validation."""
+
+import json
+
+import pytest
+import torch
+
+from distllm_amd.registry import RegistrySingleton, registry
+from distllm_amd.utils import Timer, batch_data, parse_timer_logs, retry
+
+
+def test_timer_line_roundtrip(capsys, tmp_path):
+    log = tmp_path / "t.log"
+    with Timer("computed-embeddings", "file1.jsonl", logfile=log):
+        pass
+    out = capsys.readouterr().out
+    assert "[timer] [computed-embeddings, file1.jsonl] in " in out
+    stats = parse_timer_logs(log.read_text())
+    assert len(stats) == 1
+    assert stats[0].tags == ["computed-embeddings", "file1.jsonl"]
+    assert stats[0].elapsed_s >= 0
+    assert stats[0].end_unix >= stats[0].start_unix
+
+
+def test_batch_data():
+    assert batch_data([1, 2, 3, 4, 5], 2) == [[1, 2], [3, 4], [5]]
+    assert batch_data([], 3) == []
+    with pytest.raises(ValueError):
+        batch_data([1], 0)
+
+
+def test_registry_warm_start():
+    calls = []
+
+    def factory(x):
+        calls.append(x)
+        return {"x": x}
+
+    r = RegistrySingleton()
+    a = r.get(factory, 1)
+    b = r.get(factory, 1)
+    assert a is b and calls == [1]
+    c = r.get(factory, 2)  # key change -> rebuild
+    assert c["x"] == 2 and calls == [1, 2]
+    r.clear()
+
+
+def test_retry_decorator():
+    attempts = []
+
+    @retry(tries=3, base_delay=0.01)
+    def flaky():
+        attempts.append(1)
+        if len(attempts) < 3:
+            raise RuntimeError("boom")
+        return "ok"
+
+    assert flaky() == "ok"
+    assert len(attempts) == 3
+
+
+def test_mcqa_boot_local_validation():
+    from distllm_amd.mcqa.harness import ModelConfig
+
+    with pytest.raises(Exception):
+        ModelConfig(boot_local=True)  # needs local_config_path
+    ModelConfig(boot_local=False)  # fine
+
+
+def test_tokenization_entry(tmp_path):
+    from distllm_amd.distributed_tokenization import (
+        TokenizationConfig,
+        main,
+    )
+
+    idir = tmp_path / "in"
+    idir.mkdir()
+    with open(idir / "a.jsonl", "w") as fh:
+        for i in range(4):
+            fh.write(json.dumps({"text": f"hello world {i}"}) + "\n")
+    odir = tmp_path / "out"
+    main(TokenizationConfig(input_dir=idir, output_dir=odir,
+                            save_labels=True))
+    from datasets import load_from_disk
+
+    shards = list((odir / "tokenized").iterdir())
+    assert len(shards) == 1
+    ds = load_from_disk(str(shards[0]))
+    assert set(ds.column_names) == {"input_ids", "attention_mask", "labels"}
+    assert len(ds) == 4
+
+
+def test_esm_tokenizer_exact_ids():
+    from distllm_amd.embed.tokenizers import EsmTokenizer
+
+    tok = EsmTokenizer()
+    enc = tok(["MKV"], return_tensors="pt")
+    # <cls> M K V <eos> == 0, 20, 15, 7, 2
+    assert enc["input_ids"][0].tolist()[:5] == [0, 20, 15, 7, 2]
+    assert tok.decode(enc["input_ids"][0]) == "MKV"
+
+
+def test_model_weight_loader_roundtrip(tmp_path):
+    """Native checkpoint save -> load_weights restores exactly."""
+    from safetensors.torch import save_file
+
+    from distllm_amd.models import build_model
+    from distllm_amd.models.loader import load_weights
+
+    m1 = build_model("bert-tiny", dtype=torch.float32)
+    sd = {k: v.detach().clone().contiguous()
+          for k, v in m1.named_parameters()}
+    save_file(sd, str(tmp_path / "model.safetensors"))
+    m2 = build_model("bert-tiny", dtype=torch.float32)
+    assert not torch.equal(m1.word_emb, m2.word_emb)
+    load_weights(m2, tmp_path)
+    assert torch.equal(m1.word_emb, m2.word_emb)
+    assert torch.equal(m1.layers[0].wqkv, m2.layers[0].wqkv)
