@@ -121,17 +121,30 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
           16, 0, 0);
     }
   };
+  // per-thread V staging geometry, hoisted out of the tile loop
+  int v_k[VCH];                       // key index within tile
+  const unsigned short* v_src[VCH];   // &V[v_k][d0] (tile k0 = 0)
+  int v_woff[VCH];                    // LDS elem offset d0*VROW + k
+#pragma unroll
+  for (int c = 0; c < VCH; ++c) {
+    const int ci = threadIdx.x + c * 256;
+    v_k[c] = ci / NCH;
+    const int d0 = (ci % NCH) * 8;
+    v_src[c] = Vb + (long)v_k[c] * tstr_k + d0;
+    v_woff[c] = d0 * VROW + v_k[c];
+  }
+  const unsigned short* v_src_last = Vb + (long)(SKV - 1) * tstr_k;
+
   // T14 split: issue V loads into regs (no wait) ...
   auto v_issue = [&](int k0, s16x8* vreg) {
+    const long koff = (long)k0 * tstr_k;
 #pragma unroll
     for (int c = 0; c < VCH; ++c) {
-      const int ci = threadIdx.x + c * 256;
-      const int k = ci / NCH;
-      const int d0 = (ci % NCH) * 8;
-      int krow = k0 + k;
-      if (krow >= SKV) krow = SKV - 1;
-      vreg[c] = *reinterpret_cast<const s16x8*>(
-          Vb + (long)krow * tstr_k + d0);
+      const unsigned short* src =
+          (k0 + v_k[c] < SKV)
+              ? v_src[c] + koff
+              : v_src_last + ((v_src[c] - Vb) - (long)v_k[c] * tstr_k);
+      vreg[c] = *reinterpret_cast<const s16x8*>(src);
     }
   };
   // ... and transpose-write them into the padded V^T image later.
@@ -139,12 +152,9 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     unsigned short* vt = reinterpret_cast<unsigned short*>(smem + offV[buf]);
 #pragma unroll
     for (int c = 0; c < VCH; ++c) {
-      const int ci = threadIdx.x + c * 256;
-      const int k = ci / NCH;
-      const int d0 = (ci % NCH) * 8;
 #pragma unroll
       for (int i = 0; i < 8; ++i)
-        vt[(d0 + i) * VROW + k] = (unsigned short)vreg[c][i];
+        vt[v_woff[c] + i * VROW] = (unsigned short)vreg[c][i];
     }
   };
   auto stage_m = [&](int k0, int buf) {
@@ -184,6 +194,10 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
 
     const bool wave_active =
         q0w < SQ && (!CAUSAL || k0 <= q0w + QWAVE - 1);
+    // mask values are only needed when a real mask exists or the tile
+    // hangs over the end of the sequence (padded keys); skipping the 32
+    // LDS reads otherwise removes a large per-tile VALU cost.
+    const bool tile_masked = HAS_MASK || (k0 + KVBLK > SKV);
 
     if (wave_active) {
       // ---- QK^T: two 32-key blocks --------------------------------
@@ -217,27 +231,36 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int key_loc = kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * h;
-          float s2 = accP[kb][r] * scale2 + mrow[key_loc];
+          float s2 = accP[kb][r] * scale2;
+          if (tile_masked) s2 += mrow[key_loc];
           if (CAUSAL && (k0 + key_loc > qrow_abs)) s2 = -INFINITY;
           p[kb * 16 + r] = s2;
           pmax = fmaxf(pmax, s2);
         }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
-      const float mn = fmaxf(fmaxf(m, pmax), -1e30f);
-      const float alpha = exp2f(m - mn);
+      // defer-max (guide T13, THR=8 in log2 units): skip the O rescale
+      // while the running max has not grown by more than THR — P is then
+      // bounded by 2^THR, which the fp32 l/O accumulation absorbs. The
+      // decision covers THIS tile's P entirely (textbook-safe order:
+      // nothing of this tile is exponentiated before the decision).
+      float mn = m, alpha = 1.f;
+      if (!__all(pmax - m <= 8.f)) {
+        mn = fmaxf(fmaxf(m, pmax), -1e30f);
+        alpha = __builtin_amdgcn_exp2f(m - mn);
+#pragma unroll
+        for (int d = 0; d < DBLK; ++d)
+#pragma unroll
+          for (int i = 0; i < 16; ++i) accO[d][i] *= alpha;
+      }
       float rs = 0.f;
 #pragma unroll
       for (int r = 0; r < 32; ++r) {
-        p[r] = exp2f(p[r] - mn);
+        p[r] = __builtin_amdgcn_exp2f(p[r] - mn);
         rs += p[r];
       }
       rs += __shfl_xor(rs, 32, 64);
       l = l * alpha + rs;
       m = mn;
-#pragma unroll
-      for (int d = 0; d < DBLK; ++d)
-#pragma unroll
-        for (int i = 0; i < 16; ++i) accO[d][i] *= alpha;
 
       // ---- P -> bf16 fragments (cvt + permlane32_swap) -------------
       frag_u pa[4];
