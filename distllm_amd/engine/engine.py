@@ -101,6 +101,7 @@ class LLMEngine:
             # cap: enough for max_batch_size * max_model_len
             need = (self.max_model_len // self.BLK + 1) * max_batch_size
             num_blocks = min(num_blocks, need)
+        num_blocks += 1  # block 0 is reserved (graph-padding dummy)
         self.num_blocks = num_blocks
         self.kcaches = [
             torch.zeros(num_blocks, hkv, self.BLK, hd, dtype=model.dtype,
