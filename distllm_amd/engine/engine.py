@@ -89,7 +89,8 @@ class LLMEngine:
         self.max_batch_size = max_batch_size
         self.max_model_len = min(max_model_len, d.max_position_embeddings)
         dev = self.device
-        hkv, hd, L = d.kv_heads, d.head_dim, d.num_layers
+        hkv = getattr(model, 'local_kv_heads', d.kv_heads)
+        hd, L = d.head_dim, d.num_layers
         bytes_per_block_layer = hkv * self.BLK * hd * 2 * 2  # k+v bf16
         if num_blocks is None:
             if str(dev).startswith("cuda"):
@@ -114,8 +115,12 @@ class LLMEngine:
         # hipGraph-captured decode (one graph per pow2 batch bucket):
         # the decode step is ~300 small launches per step for a 32-layer
         # model — graph replay removes the launch overhead.
+        tp = getattr(model, 'tp', 1)
         self.use_graphs = (use_graphs if use_graphs is not None
-                           else str(dev).startswith("cuda"))
+                           else (str(dev).startswith("cuda") and tp == 1))
+        # deterministic sampling RNG shared across TP ranks
+        self._rng = torch.Generator(device=dev)
+        self._rng.manual_seed(0x5EED)
         self.max_blocks_per_seq = (self.max_model_len // self.BLK) + 1
         self._graphs: dict[int, "_GraphRunner"] = {}
 
@@ -241,7 +246,8 @@ class LLMEngine:
         if params.top_p < 1.0:
             next_ids = self._sample_top_p(logits, params)
         else:
-            rand = torch.rand(len(seqs), device=dev)
+            rand = torch.rand(len(seqs), device=dev,
+                              generator=self._rng)
             next_ids = ops.sample_tokens(logits, rand, params.temperature,
                                          params.min_p)
         next_list = next_ids.tolist()

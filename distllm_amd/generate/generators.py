@@ -54,7 +54,22 @@ class NativeGenerator:
         dims = get_dims(preset)
         device = "cuda" if torch.cuda.is_available() else "cpu"
         dtype = torch.bfloat16 if device != "cpu" else torch.float32
-        self.model = build_model(preset, device=device, dtype=dtype)
+        kw = {}
+        if config.tensor_parallel_size > 1:
+            import torch.distributed as dist
+
+            if not dist.is_initialized():
+                raise RuntimeError(
+                    "tensor_parallel_size > 1 requires torch.distributed "
+                    "(launch under torchrun with one rank per GPU)"
+                )
+            if dist.get_world_size() != config.tensor_parallel_size:
+                raise RuntimeError(
+                    "tensor_parallel_size must equal the world size "
+                    "(TP subgroups are not split out in this round)"
+                )
+            kw["tp_group"] = dist.group.WORLD
+        self.model = build_model(preset, device=device, dtype=dtype, **kw)
         if config.weights_path:
             from ..models.loader import load_weights
 

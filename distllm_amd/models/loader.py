@@ -63,18 +63,39 @@ def _map_llama(src: dict, model) -> dict:
     out["final_norm"] = src[f"{p}norm.weight"]
     if getattr(model, "lm_head", None) is not None and "lm_head.weight" in src:
         out["lm_head"] = src["lm_head.weight"]
+    # tensor-parallel shard slicing (Megatron-style)
+    tp = getattr(model, "tp", 1)
+    r = getattr(model, "tp_rank", 0)
+    d = model.dims
+    hd = d.head_dim
+
+    def rows(w, total):
+        sh = total // tp
+        return w[r * sh : (r + 1) * sh]
+
+    def cols(w, total):
+        sh = total // tp
+        return w[:, r * sh : (r + 1) * sh]
+
     for i in range(len(model.layers)):
         lp = f"{p}layers.{i}."
-        out[f"layers.{i}.wq"] = src[lp + "self_attn.q_proj.weight"]
-        out[f"layers.{i}.wk"] = src[lp + "self_attn.k_proj.weight"]
-        out[f"layers.{i}.wv"] = src[lp + "self_attn.v_proj.weight"]
-        out[f"layers.{i}.wo"] = src[lp + "self_attn.o_proj.weight"]
+        out[f"layers.{i}.wq"] = rows(src[lp + "self_attn.q_proj.weight"],
+                                     d.num_heads * hd)
+        out[f"layers.{i}.wk"] = rows(src[lp + "self_attn.k_proj.weight"],
+                                     d.kv_heads * hd)
+        out[f"layers.{i}.wv"] = rows(src[lp + "self_attn.v_proj.weight"],
+                                     d.kv_heads * hd)
+        out[f"layers.{i}.wo"] = cols(src[lp + "self_attn.o_proj.weight"],
+                                     d.num_heads * hd)
         out[f"layers.{i}.in_norm"] = src[lp + "input_layernorm.weight"]
         out[f"layers.{i}.post_norm"] = src[
             lp + "post_attention_layernorm.weight"]
-        out[f"layers.{i}.wgate"] = src[lp + "mlp.gate_proj.weight"]
-        out[f"layers.{i}.wup"] = src[lp + "mlp.up_proj.weight"]
-        out[f"layers.{i}.wdown"] = src[lp + "mlp.down_proj.weight"]
+        out[f"layers.{i}.wgate"] = rows(src[lp + "mlp.gate_proj.weight"],
+                                        d.intermediate_size)
+        out[f"layers.{i}.wup"] = rows(src[lp + "mlp.up_proj.weight"],
+                                      d.intermediate_size)
+        out[f"layers.{i}.wdown"] = cols(src[lp + "mlp.down_proj.weight"],
+                                        d.intermediate_size)
     return out
 
 

@@ -24,34 +24,52 @@ from .config import ModelDims
 
 
 class DecoderLayer(nn.Module):
-    def __init__(self, d: ModelDims):
+    def __init__(self, d: ModelDims, tp: int = 1):
         super().__init__()
         H = d.hidden_size
         self.d = d
-        kvh = d.kv_heads
+        kvh = d.kv_heads // tp
+        nh = d.num_heads // tp
         hd = d.head_dim
-        self.wq = nn.Parameter(torch.empty(H, H))
+        inter = d.intermediate_size // tp
+        # column-parallel q/k/v & gate/up, row-parallel wo & wdown
+        # (Megatron-style sharding; all-reduce after wo and wdown)
+        self.wq = nn.Parameter(torch.empty(nh * hd, H))
         self.wk = nn.Parameter(torch.empty(kvh * hd, H))
         self.wv = nn.Parameter(torch.empty(kvh * hd, H))
-        self.wo = nn.Parameter(torch.empty(H, H))
+        self.wo = nn.Parameter(torch.empty(H, nh * hd))
         self.in_norm = nn.Parameter(torch.empty(H))
         self.post_norm = nn.Parameter(torch.empty(H))
-        self.wgate = nn.Parameter(torch.empty(d.intermediate_size, H))
-        self.wup = nn.Parameter(torch.empty(d.intermediate_size, H))
-        self.wdown = nn.Parameter(torch.empty(H, d.intermediate_size))
+        self.wgate = nn.Parameter(torch.empty(inter, H))
+        self.wup = nn.Parameter(torch.empty(inter, H))
+        self.wdown = nn.Parameter(torch.empty(H, inter))
 
 
 class DecoderModel(nn.Module):
     """forward() -> last_hidden_state; generation via prefill()/decode()."""
 
     def __init__(self, dims: ModelDims, dtype=torch.bfloat16,
-                 with_lm_head: bool | None = None):
+                 with_lm_head: bool | None = None,
+                 tp_group=None):
         super().__init__()
+        import torch.distributed as dist
+
         self.dims = dims
+        self.tp_group = tp_group
+        self.tp = (dist.get_world_size(tp_group)
+                   if tp_group is not None and dist.is_initialized() else 1)
+        self.tp_rank = (dist.get_rank(tp_group)
+                        if tp_group is not None and dist.is_initialized()
+                        else 0)
+        if dims.num_heads % self.tp or dims.kv_heads % self.tp or \
+                dims.intermediate_size % self.tp:
+            raise ValueError(
+                f"tp={self.tp} must divide heads/kv_heads/intermediate"
+            )
         H = dims.hidden_size
         self.tok_emb = nn.Parameter(torch.empty(dims.vocab_size, H))
         self.layers = nn.ModuleList(
-            DecoderLayer(dims) for _ in range(dims.num_layers)
+            DecoderLayer(dims, tp=self.tp) for _ in range(dims.num_layers)
         )
         self.final_norm = nn.Parameter(torch.empty(H))
         if with_lm_head is None:
@@ -95,10 +113,25 @@ class DecoderModel(nn.Module):
             self._rope_cache = (cs, sn)
         return self._rope_cache
 
+    @property
+    def local_heads(self) -> int:
+        return self.dims.num_heads // self.tp
+
+    @property
+    def local_kv_heads(self) -> int:
+        return self.dims.kv_heads // self.tp
+
+    def _allreduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.tp > 1:
+            import torch.distributed as dist
+
+            dist.all_reduce(x, group=self.tp_group)
+        return x
+
     def _attn_block(self, layer: DecoderLayer, x, mask_f, pos, kv_out=None):
         d = self.dims
         B, S, H = x.shape
-        nh, kvh, hd = d.num_heads, d.kv_heads, d.head_dim
+        nh, kvh, hd = self.local_heads, self.local_kv_heads, d.head_dim
         cs, sn = self._rope(x.device)
         h = ops.rms_norm(x, layer.in_norm, eps=d.rms_norm_eps)
         q = ops.linear(h, layer.wq).view(B, S, nh, hd)
@@ -110,12 +143,13 @@ class DecoderModel(nn.Module):
             kv_out.append((k, v))
         ctx = ops.attn_fwd(q, k, v, mask=mask_f, causal=True,
                            scale=1.0 / math.sqrt(hd))
-        attn = ops.linear(ctx.reshape(B, S, H), layer.wo)
-        x = x + attn
+        attn = ops.linear(ctx.reshape(B, S, nh * hd), layer.wo)
+        x = x + self._allreduce(attn)
         h = ops.rms_norm(x, layer.post_norm, eps=d.rms_norm_eps)
         gate = ops.linear(h, layer.wgate)
         up = ops.linear(h, layer.wup)
-        x = x + ops.linear(ops.swiglu(gate, up), layer.wdown)
+        down = ops.linear(ops.swiglu(gate, up), layer.wdown)
+        x = x + self._allreduce(down)
         return x
 
     @torch.no_grad()
@@ -164,7 +198,7 @@ class DecoderModel(nn.Module):
         """
         d = self.dims
         B = tokens.shape[0]
-        nh, kvh, hd = d.num_heads, d.kv_heads, d.head_dim
+        nh, kvh, hd = self.local_heads, self.local_kv_heads, d.head_dim
         cs, sn = self._rope(tokens.device)
         x = self.tok_emb[tokens.long()].unsqueeze(1)  # [B,1,H]
         pos = (ctx_lens - 1).int()
@@ -181,11 +215,12 @@ class DecoderModel(nn.Module):
             ctx = ops.paged_attn(q[:, 0].contiguous(), kcaches[li],
                                  vcaches[li], block_tables, ctx_lens,
                                  scale=1.0 / math.sqrt(hd))
-            attn = ops.linear(ctx.reshape(B, 1, d.hidden_size), layer.wo)
-            x = x + attn
+            attn = ops.linear(ctx.reshape(B, 1, nh * hd), layer.wo)
+            x = x + self._allreduce(attn)
             h = ops.rms_norm(x, layer.post_norm, eps=d.rms_norm_eps)
             gate = ops.linear(h, layer.wgate)
             up = ops.linear(h, layer.wup)
-            x = x + ops.linear(ops.swiglu(gate, up), layer.wdown)
+            down = ops.linear(ops.swiglu(gate, up), layer.wdown)
+            x = x + self._allreduce(down)
         x = ops.rms_norm(x, self.final_norm, eps=d.rms_norm_eps)
         return self.logits(x[:, 0])

@@ -80,3 +80,11 @@ def test_broadcast_model_helper():
     w0 = m.word_emb.clone()
     broadcast_model(m)
     assert torch.equal(w0, m.word_emb)
+
+
+def test_tensor_parallel_two_ranks(tmp_path):
+    """TP=2 decoder forward + engine decode match the single-rank model
+    exactly (fp32, gloo)."""
+    r = _torchrun(["tests/tp_worker.py", str(tmp_path)], timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "TP_OK" in r.stdout, r.stdout
