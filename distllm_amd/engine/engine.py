@@ -171,14 +171,14 @@ class LLMEngine:
 
     def _block_table(self, seqs: list[_Seq],
                      width: int | None = None) -> torch.Tensor:
+        # build on the host, upload in ONE copy (a per-row device
+        # tensor-from-list costs one tiny H2D copy per sequence/step)
         maxb = width or max(len(s.blocks) for s in seqs)
-        bt = torch.zeros(len(seqs), maxb, dtype=torch.int32,
-                         device=self.device)
+        bt = torch.zeros(len(seqs), maxb, dtype=torch.int32)
         for i, s in enumerate(seqs):
             bt[i, : len(s.blocks)] = torch.tensor(s.blocks,
-                                                  dtype=torch.int32,
-                                                  device=self.device)
-        return bt
+                                                  dtype=torch.int32)
+        return bt.to(self.device, non_blocking=True)
 
     @torch.no_grad()
     def _run_prefill(self, seqs: list[_Seq], params: SamplingParams,

@@ -128,12 +128,16 @@ class DecoderModel(nn.Module):
             dist.all_reduce(x, group=self.tp_group)
         return x
 
-    def _attn_block(self, layer: DecoderLayer, x, mask_f, pos, kv_out=None):
+    def _attn_block(self, layer: DecoderLayer, h, x, mask_f, pos,
+                    kv_out=None):
+        """One decoder layer given h = rms_norm(x, in_norm) already
+        computed (residual adds are fused into the norm kernels).
+        Returns (mlp_out, x) where the caller fuses mlp_out + x into
+        the next norm."""
         d = self.dims
-        B, S, H = x.shape
+        B, S, _ = h.shape
         nh, kvh, hd = self.local_heads, self.local_kv_heads, d.head_dim
-        cs, sn = self._rope(x.device)
-        h = ops.rms_norm(x, layer.in_norm, eps=d.rms_norm_eps)
+        cs, sn = self._rope(h.device)
         q = ops.linear(h, layer.wq).view(B, S, nh, hd)
         k = ops.linear(h, layer.wk).view(B, S, kvh, hd)
         v = ops.linear(h, layer.wv).view(B, S, kvh, hd)
@@ -143,14 +147,16 @@ class DecoderModel(nn.Module):
             kv_out.append((k, v))
         ctx = ops.attn_fwd(q, k, v, mask=mask_f, causal=True,
                            scale=1.0 / math.sqrt(hd))
-        attn = ops.linear(ctx.reshape(B, S, nh * hd), layer.wo)
-        x = x + self._allreduce(attn)
-        h = ops.rms_norm(x, layer.post_norm, eps=d.rms_norm_eps)
-        gate = ops.linear(h, layer.wgate)
-        up = ops.linear(h, layer.wup)
-        down = ops.linear(ops.swiglu(gate, up), layer.wdown)
-        x = x + self._allreduce(down)
-        return x
+        attn = self._allreduce(
+            ops.linear(ctx.reshape(B, S, nh * hd), layer.wo))
+        # fused: x' = x + attn ; h2 = rms_norm(x')
+        h2, x = ops.norm_fwd_sum(attn, x, layer.post_norm,
+                                 eps=d.rms_norm_eps, rms=True)
+        gate = ops.linear(h2, layer.wgate)
+        up = ops.linear(h2, layer.wup)
+        down = self._allreduce(
+            ops.linear(ops.swiglu(gate, up), layer.wdown))
+        return down, x
 
     @torch.no_grad()
     def forward(self, input_ids: torch.Tensor, attention_mask: torch.Tensor,
@@ -169,9 +175,15 @@ class DecoderModel(nn.Module):
             .repeat(B, 1)
             .reshape(-1)
         )
-        for layer in self.layers:
-            x = self._attn_block(layer, x, mask_f, pos, kv_out)
-        return ops.rms_norm(x, self.final_norm, eps=self.dims.rms_norm_eps)
+        d = self.dims
+        h = ops.rms_norm(x, self.layers[0].in_norm, eps=d.rms_norm_eps)
+        for li, layer in enumerate(self.layers):
+            mlp_out, x = self._attn_block(layer, h, x, mask_f, pos, kv_out)
+            next_w = (self.layers[li + 1].in_norm
+                      if li + 1 < len(self.layers) else self.final_norm)
+            h, x = ops.norm_fwd_sum(mlp_out, x, next_w,
+                                    eps=d.rms_norm_eps, rms=True)
+        return h
 
     @torch.no_grad()
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
@@ -203,8 +215,8 @@ class DecoderModel(nn.Module):
         x = self.tok_emb[tokens.long()].unsqueeze(1)  # [B,1,H]
         pos = (ctx_lens - 1).int()
         slot = pos
+        h = ops.rms_norm(x, self.layers[0].in_norm, eps=d.rms_norm_eps)
         for li, layer in enumerate(self.layers):
-            h = ops.rms_norm(x, layer.in_norm, eps=d.rms_norm_eps)
             q = ops.linear(h, layer.wq).view(B, 1, nh, hd)
             k = ops.linear(h, layer.wk).view(B, 1, kvh, hd)
             v = ops.linear(h, layer.wv).view(B, 1, kvh, hd)
@@ -215,12 +227,16 @@ class DecoderModel(nn.Module):
             ctx = ops.paged_attn(q[:, 0].contiguous(), kcaches[li],
                                  vcaches[li], block_tables, ctx_lens,
                                  scale=1.0 / math.sqrt(hd))
-            attn = ops.linear(ctx.reshape(B, 1, nh * hd), layer.wo)
-            x = x + self._allreduce(attn)
-            h = ops.rms_norm(x, layer.post_norm, eps=d.rms_norm_eps)
-            gate = ops.linear(h, layer.wgate)
-            up = ops.linear(h, layer.wup)
-            down = ops.linear(ops.swiglu(gate, up), layer.wdown)
-            x = x + self._allreduce(down)
-        x = ops.rms_norm(x, self.final_norm, eps=d.rms_norm_eps)
-        return self.logits(x[:, 0])
+            attn = self._allreduce(
+                ops.linear(ctx.reshape(B, 1, nh * hd), layer.wo))
+            h2, x = ops.norm_fwd_sum(attn, x, layer.post_norm,
+                                     eps=d.rms_norm_eps, rms=True)
+            gate = ops.linear(h2, layer.wgate)
+            up = ops.linear(h2, layer.wup)
+            down = self._allreduce(
+                ops.linear(ops.swiglu(gate, up), layer.wdown))
+            next_w = (self.layers[li + 1].in_norm
+                      if li + 1 < len(self.layers) else self.final_norm)
+            h, x = ops.norm_fwd_sum(down, x, next_w,
+                                    eps=d.rms_norm_eps, rms=True)
+        return self.logits(h[:, 0])

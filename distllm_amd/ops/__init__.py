@@ -48,6 +48,17 @@ def norm_fwd(x, res=None, w=None, b=None, eps=1e-5, rms=False):
     return torch_ref.norm_fwd(x, res, w, b, eps, rms)
 
 
+def norm_fwd_sum(x, res, w, b=None, eps=1e-5, rms=False):
+    """Fused (x+res) -> norm; returns (normed, x+res). Replaces the
+    separate residual-add elementwise pass in pre-norm decoders."""
+    if x.is_cuda:
+        _require_hip()
+        summed = torch.empty_like(x)
+        out = _hip_ops.norm_fwd(x, res, w, b, summed, eps, rms)
+        return out, summed
+    return torch_ref.norm_fwd(x, res, w, b, eps, rms, return_sum=True)
+
+
 def layer_norm(x, w, b, eps=1e-12, res=None):
     return norm_fwd(x, res=res, w=w, b=b, eps=eps, rms=False)
 
@@ -118,35 +129,74 @@ def gemm(A, B, bias=None, act=ACT_NONE):
     return torch_ref.gemm(A, B, bias, act)
 
 
+_GEMM_CHOICE: dict[tuple, str] = {}
+
+
+def _blaslt_path(x2, weight, bias, act):
+    out = torch.nn.functional.linear(x2, weight, bias)
+    if act == ACT_GELU_TANH:
+        out = bias_gelu(out, None, erf=False)
+    elif act == ACT_GELU_ERF:
+        out = bias_gelu(out, None, erf=True)
+    elif act == ACT_SILU:
+        out = (
+            torch.nn.functional.silu(out.float()).to(out.dtype)
+            if not out.is_cuda else _silu_gpu(out)
+        )
+    return out
+
+
+def _autotune_gemm(x2, weight, bias, act):
+    """Measure hipBLASLt(+epilogue kernel) vs the fused gemm8 MFMA
+    kernel once per (M-bucket, N, K, act) and cache the winner."""
+    import time as _t
+
+    def timeit(fn):
+        fn()
+        torch.cuda.synchronize()
+        t0 = _t.perf_counter()
+        for _ in range(5):
+            fn()
+        torch.cuda.synchronize()
+        return _t.perf_counter() - t0
+
+    t_blas = timeit(lambda: _blaslt_path(x2, weight, bias, act))
+    t_hip = timeit(lambda: _hip_ops.gemm8_bf16(x2, weight, bias, act))
+    return "gemm8" if t_hip < t_blas else "blaslt"
+
+
 def linear(x, weight, bias=None, act=ACT_NONE):
-    """Linear layer entry used by models: hipBLASLt (torch.matmul) for
-    the plain GEMM + fused HIP epilogue, or the hand-written fused MFMA
-    GEMM when DISTLLM_GEMM=hip and shapes allow."""
+    """Linear layer entry used by models. GPU path: autotuned choice
+    between hipBLASLt (+fused activation kernel) and the hand-written
+    deep-pipelined MFMA GEMM with fused bias+activation epilogue
+    (gemm8.hip). Override with DISTLLM_GEMM=blaslt|gemm8."""
     import os
 
     orig_shape = x.shape
     x2 = x.reshape(-1, orig_shape[-1])
     M, K = x2.shape
     N = weight.shape[0]
-    use_hip_gemm = (
-        x.is_cuda
-        and os.environ.get("DISTLLM_GEMM", "blaslt") == "hip"
-        and M % 128 == 0 and N % 128 == 0 and K % 64 == 0
-    )
-    if use_hip_gemm:
-        out = gemm(x2.contiguous(), weight, bias, act)
+    if x.is_cuda:
+        _require_hip()
+        mode = os.environ.get("DISTLLM_GEMM", "auto")
+        shapes_ok = (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
+                     and x2.is_contiguous())
+        if shapes_ok and mode != "blaslt":
+            if mode == "gemm8":
+                out = _hip_ops.gemm8_bf16(x2, weight, bias, act)
+            else:  # auto
+                key = (min(M, 1 << 20) // 4096, N, K, act)
+                choice = _GEMM_CHOICE.get(key)
+                if choice is None:
+                    choice = _autotune_gemm(x2, weight, bias, act)
+                    _GEMM_CHOICE[key] = choice
+                out = (_hip_ops.gemm8_bf16(x2, weight, bias, act)
+                       if choice == "gemm8"
+                       else _blaslt_path(x2, weight, bias, act))
+        else:
+            out = _blaslt_path(x2, weight, bias, act)
     else:
-        out = torch.nn.functional.linear(x2, weight, bias)
-        if act == ACT_GELU_TANH:
-            out = bias_gelu(out, None, erf=False)
-        elif act == ACT_GELU_ERF:
-            out = bias_gelu(out, None, erf=True)
-        elif act == ACT_SILU:
-            out = (
-                torch.nn.functional.silu(out.float()).to(out.dtype)
-                if not out.is_cuda
-                else _silu_gpu(out)
-            )
+        out = _blaslt_path(x2, weight, bias, act)
     return out.reshape(*orig_shape[:-1], N)
 
 

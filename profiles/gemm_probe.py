@@ -53,14 +53,22 @@ def main():
         def hip():
             return ops.gemm(A, B, bias, act=act)
 
+        def hip8():
+            return ops._hip_ops.gemm8_bf16(A, B, bias, act)
+
         t1 = bench(blaslt)
         t2 = bench(hip)
+        t3 = bench(hip8)
+        r8 = hip8()[:256]
+        ref8 = ops.torch_ref.gemm(A[:256].float().bfloat16(), B, bias, act)
+        err8 = (r8.float() - ref8.float()).abs().max().item()
         r1 = ops.torch_ref.gemm(A[:256].float().bfloat16(), B, bias, act)
         r2 = hip()[:256]
         err = (r2.float() - r1.float()).abs().max().item()
-        print(f"{name:<10} M{m} N{n} K{k}: blaslt+epi {t1 * 1e3:7.3f} ms "
-              f"({tf / t1:6.0f} TF) | hip fused {t2 * 1e3:7.3f} ms "
-              f"({tf / t2:6.0f} TF) | maxerr {err:.3f}")
+        print(f"{name:<10} M{m} N{n} K{k}: blaslt {t1 * 1e3:7.3f} ms "
+              f"({tf / t1:6.0f} TF) | g1 {t2 * 1e3:7.3f} ({tf / t2:5.0f} TF) "
+              f"| g8 {t3 * 1e3:7.3f} ({tf / t3:5.0f} TF) "
+              f"| err {err:.3f}/{err8:.3f}")
 
 
 if __name__ == "__main__":
