@@ -3,13 +3,14 @@
 // vLLM, generate/generators/vllm_backend.py).
 //
 // Cache layout: kcache/vcache [nblocks, HKV, BLK, D] bf16, BLK=16.
-// block_table [B, max_blocks] int32 ; ctx_len [B] int32 (tokens in ctx,
-// including the one being attended as the newest position).
+// block_table [B, max_blocks] int32 ; ctx_len [B] int32.
 // q: [B, H, D] bf16 ; out: [B, H, D] bf16.
 //
-// Decode is KV-bandwidth-bound: one workgroup per (b, h); 4 waves split
-// the context; lane-per-key dot for QK, d-sliced coalesced accumulate
-// for PV through an LDS p-buffer; cross-wave combine via LDS (m, l, acc).
+// Decode is latency-bound at small contexts: the v2 kernel stages the
+// block table in LDS once, keeps the QK dot lane-per-key (independent
+// vector loads), and runs the PV accumulation branch-free with an
+// 8-deep unroll so the per-key V reads pipeline instead of serializing
+// on L2 latency. Cross-wave combine via LDS.
 #include "common.h"
 
 #define PBLK 16  // tokens per KV block
@@ -20,8 +21,9 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
                   const unsigned short* __restrict__ kcache,
                   const unsigned short* __restrict__ vcache,
                   const int* __restrict__ block_table,
-                  const int* __restrict__ ctx_len, unsigned short* __restrict__ out,
-                  int B, int H, int HKV, int max_blocks, float scale) {
+                  const int* __restrict__ ctx_len,
+                  unsigned short* __restrict__ out, int B, int H, int HKV,
+                  int max_blocks, float scale) {
   const int b = blockIdx.x / H;
   const int h = blockIdx.x % H;
   const int kvh = h / (H / HKV);
@@ -30,31 +32,32 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
   const int L = ctx_len[b];
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: p_lds [4][64] f32 | q_lds [D] f32 | wave state m,l [4][2]
-  float* p_lds = reinterpret_cast<float*>(smem);                 // 1 KiB
-  float* q_lds = p_lds + 4 * 64;                                 // D f32
-  float* wstate = q_lds + D;                                     // 8 f32
-  float* wacc = wstate + 8;                                      // 4*D f32
+  float* p_lds = reinterpret_cast<float*>(smem);                 // 4*64
+  float* q_lds = p_lds + 4 * 64;                                 // D
+  float* wstate = q_lds + D;                                     // 8
+  float* wacc = wstate + 8;                                      // 4*D
+  int* bt_lds = reinterpret_cast<int*>(wacc + 4 * D);            // nblk
 
-  // load q into LDS as f32 (scaled)
+  const int nblk = (L + PBLK - 1) / PBLK;
+  for (int i = threadIdx.x; i < nblk; i += 256)
+    bt_lds[i] = block_table[(long)b * max_blocks + i];
   for (int d = threadIdx.x; d < D; d += 256)
     q_lds[d] = bf2f(q[((long)b * H + h) * D + d]) * scale;
   __syncthreads();
 
-  // each wave processes keys [wid*64 + 256*t ...] lane-per-key
   float m = -1e30f, l = 0.f;
-  float acc[2] = {0.f, 0.f};  // d-slice accumulator: lane owns D/64 dims
-  constexpr int DPL = D / 64; // 1 (D=64) or 2 (D=128)
-
+  float acc[2] = {0.f, 0.f};
+  constexpr int DPL = D / 64;  // d-cols per lane (1 or 2)
   const long blk_stride = (long)HKV * PBLK * D;
+  const long head_off = (long)kvh * PBLK * D;
 
   for (int t0 = wid * 64; t0 < L; t0 += 256) {
     const int key = t0 + lane;
     float s = -INFINITY;
     if (key < L) {
-      const int blk = block_table[(long)b * max_blocks + key / PBLK];
-      const unsigned short* krow =
-          kcache + (long)blk * blk_stride + ((long)kvh * PBLK + key % PBLK) * D;
+      const unsigned short* krow = kcache +
+          (long)bt_lds[key / PBLK] * blk_stride + head_off +
+          (key % PBLK) * D;
       float dot = 0.f;
 #pragma unroll
       for (int c = 0; c < D / 8; ++c) {
@@ -64,39 +67,50 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
 #pragma unroll
         for (int i = 0; i < 8; ++i) dot += kf[i] * q_lds[c * 8 + i];
       }
-      s = dot;  // q already scaled
+      s = dot;
     }
-    // wave online softmax
-    float pmax = wave_reduce_max(s);
-    float mn = fmaxf(fmaxf(m, pmax), -1e30f);
-    float alpha = exp2f((m - mn) * 1.4426950408889634f);
-    float p = (key < L) ? exp2f((s - mn) * 1.4426950408889634f) : 0.f;
-    float rs = wave_reduce_sum(p);
-    l = l * alpha + rs;
+    const float pmax = wave_reduce_max(s);
+    const float mn = fmaxf(fmaxf(m, pmax), -1e30f);
+    const float alpha = exp2f((m - mn) * 1.4426950408889634f);
+    const float p = (key < L)
+        ? exp2f((s - mn) * 1.4426950408889634f) : 0.f;
+    l = l * alpha + wave_reduce_sum(p);
     m = mn;
     p_lds[wid * 64 + lane] = p;
-    // rescale accumulator
 #pragma unroll
     for (int i = 0; i < DPL; ++i) acc[i] *= alpha;
-    // PV: lane owns dims d = lane*DPL + i ... use coalesced V reads:
-    // iterate the 64 keys of this wave's tile; lane reads V[key][d_own].
-    // (wave-internal LDS p broadcast; no cross-wave sync needed)
-#pragma unroll 4
-    for (int kk = 0; kk < 64; ++kk) {
-      const int key2 = t0 + kk;
-      if (key2 >= L) break;
-      const float pv = p_lds[wid * 64 + kk];
-      if (pv == 0.f) continue;
-      const int blk = block_table[(long)b * max_blocks + key2 / PBLK];
-      const unsigned short* vrow =
-          vcache + (long)blk * blk_stride + ((long)kvh * PBLK + key2 % PBLK) * D;
+    // branch-free 8-deep PV: independent loads pipeline over L2 latency
+    const int kmax = min(64, L - t0);
+    int kk = 0;
+    for (; kk + 8 <= kmax; kk += 8) {
+      float pv[8];
+      float vv[8][DPL];
 #pragma unroll
-      for (int i = 0; i < DPL; ++i)
-        acc[i] += pv * bf2f(vrow[lane * DPL + i]);
+      for (int j = 0; j < 8; ++j) {
+        const int key2 = t0 + kk + j;
+        pv[j] = p_lds[wid * 64 + kk + j];
+        const unsigned short* vrow = vcache +
+            (long)bt_lds[key2 / PBLK] * blk_stride + head_off +
+            (key2 % PBLK) * D + lane * DPL;
+#pragma unroll
+        for (int i = 0; i < DPL; ++i) vv[j][i] = bf2f(vrow[i]);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+#pragma unroll
+        for (int i = 0; i < DPL; ++i) acc[i] += pv[j] * vv[j][i];
+    }
+    for (; kk < kmax; ++kk) {
+      const int key2 = t0 + kk;
+      const float pv = p_lds[wid * 64 + kk];
+      const unsigned short* vrow = vcache +
+          (long)bt_lds[key2 / PBLK] * blk_stride + head_off +
+          (key2 % PBLK) * D + lane * DPL;
+#pragma unroll
+      for (int i = 0; i < DPL; ++i) acc[i] += pv * bf2f(vrow[i]);
     }
   }
 
-  // cross-wave combine through LDS
   if (lane == 0) {
     wstate[wid * 2] = m;
     wstate[wid * 2 + 1] = l;
@@ -130,7 +144,7 @@ extern "C" void launch_paged_attn(const void* q, const void* kcache,
                                   const int* ctx_len, void* out, int B, int H,
                                   int HKV, int D, int max_blocks, float scale,
                                   hipStream_t stream) {
-  const int lds = (4 * 64 + D + 8 + 4 * D) * 4;
+  const int lds = (4 * 64 + D + 8 + 4 * D) * 4 + max_blocks * 4;
   dim3 grid(B * H);
   if (D == 64)
     hipLaunchKernelGGL((paged_attn_kernel<64>), grid, dim3(256), lds, stream,
