@@ -1,23 +1,25 @@
 // Deep-pipelined CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T (+bias/act).
 //
-// 256x256x64 tile, 512 threads = 8 waves (2M x 4N), each wave a 128x64
-// C tile as 8x4 fragments of mfma_f32_16x16x32_bf16. Structure follows
-// the guide's 256^2 phased template family (§5 "8-phase"), adapted to a
-// schedule whose staging-overwrite and wait placement are provable:
-//  - double-buffered A/B LDS images ([2 halves][128][64] bf16 each,
-//    XOR-swizzled chunk ^ (row&7) against ds_read_b128 conflicts;
-//    swizzle applied on the glds SOURCE address, rule 21);
-//  - 4 phases per K-tile: each phase computes one C-quadrant (64x32,
-//    16 MFMA) and stages ONE half-tile of the NEXT K-tile into the
-//    other buffer (2 glds/wave);
-//  - raw s_barrier between phases (no vmcnt(0) drain: the only VM wait
-//    is ONE counted vmcnt(2) per K-tile, placed after the first phase's
-//    stage issue — it covers all of the tile's own halves and leaves
-//    the newest half-tile in flight);
-//  - s_setprio(1) around each MFMA cluster (T5);
-//  - epilogue: per-quadrant LDS transpose rounds -> fused bias +
-//    activation -> coalesced bf16 stores.
-// Requires M%256==0, N%256==0, K%64==0 (host pads / falls back).
+// 256x256x64 tile, 512 threads = 8 waves, each wave computes ALL 256
+// rows x 32 cols (16x2 fragments of mfma_f32_16x16x32_bf16). This
+// wave->column mapping is what makes a >1-tile-deep prefetch provable
+// with double buffering: the A row-half an entire phase-pair stops
+// reading frees mid-tile, so A halves stage TWO tiles ahead and the
+// inter-tile wait is a counted vmcnt, never a drain.
+//
+// Schedule per K-tile t (cur = buf[t&1], oth = cur^1; guide §5
+// phased-template family, own derivation):
+//   ph0: ds_read A-rh0(16) + B-all(4); stage oth.A1<-A1(t+1),
+//        oth.B0<-B0(t+1); 16 MFMA (rh0, ch0);  [mid barrier]
+//   ph1: stage oth.B1<-B1(t+1), cur.A0<-A0(t+2); 16 MFMA (rh0, ch1);
+//   ph2: ds_read A-rh1(16); 16 MFMA (rh1, ch0);
+//   ph3: 16 MFMA (rh1, ch1); s_waitcnt vmcnt(2) (leaves only A0(t+2)
+//        in flight);  [boundary barrier]
+// XOR swizzle chunk^(row&7) on the glds SOURCE address (rule 21);
+// s_setprio(1) around MFMA clusters; raw s_barrier (one compiler
+// memory fence per barrier, no vmcnt drains); fused bias+activation
+// epilogue through padded-LDS transpose rounds.
+// Requires M%256==0, N%256==0, K%64==0 (host falls back otherwise).
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
@@ -26,14 +28,12 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 #define G8_BN 256
 #define G8_BK 64
 #define HTILE_B (128 * G8_BK * 2)  // one half image: 16 KiB
-// LDS: A[2buf][2half] + B[2buf][2half] = 8 * 16 KiB = 128 KiB
-#define G8_LDS (8 * HTILE_B)
+#define G8_LDS (8 * HTILE_B)       // A/B x 2 buf x 2 half = 128 KiB
 
 extern __shared__ __attribute__((aligned(16))) char smem[];
 
 enum G8Act { G8_NONE = 0, G8_GELU_TANH = 1, G8_GELU_ERF = 2, G8_SILU = 3 };
 
-// image byte offset for element (row 0..127, chunk16 0..7) with swizzle
 __device__ __forceinline__ int g8_img_off(int row, int c16) {
   return row * 128 + ((c16 ^ (row & 7)) * 16);
 }
@@ -45,9 +45,7 @@ gemm8_kernel(const unsigned short* __restrict__ A,
              const unsigned short* __restrict__ bias,
              unsigned short* __restrict__ C, int M, int N, int K) {
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;   // 0..7
-  const int wm = wid >> 2;            // 0..1 (M half)
-  const int wn = wid & 3;             // 0..3 (N quarter)
+  const int wid = threadIdx.x >> 6;   // 0..7: wave's 32-col slice
 
   // T1 bijective XCD swizzle
   const int ntm = M / G8_BM, ntn = N / G8_BN;
@@ -63,21 +61,19 @@ gemm8_kernel(const unsigned short* __restrict__ A,
   const unsigned short* Atile = A + (long)tm * G8_BM * K;
   const unsigned short* Btile = B + (long)tn * G8_BN * K;
 
-  // LDS half images: [op][buf][half]
   auto img = [&](int op, int buf, int half) {
     return smem + ((op * 2 + buf) * 2 + half) * HTILE_B;
   };
 
-  // ---- staging: one half-tile (128 rows x 64 cols) by all 8 waves ----
-  // piece = 8 rows (1 KiB glds); wave w takes pieces 2w, 2w+1.
-  const int r_in = lane >> 3;          // row within piece
-  const int c_dst = lane & 7;          // dest chunk16
+  // stage one 128x64 half (16 glds; 2 per wave), swizzled source
+  const int r_in = lane >> 3;
+  const int c_dst = lane & 7;
   auto stage_half = [&](const unsigned short* G, int k0, int op, int buf,
                         int half) {
 #pragma unroll
     for (int pp = 0; pp < 2; ++pp) {
       const int piece = wid * 2 + pp;
-      const int row = piece * 8 + r_in;          // 0..127 within half
+      const int row = piece * 8 + r_in;
       const int c_src = c_dst ^ (row & 7);
       const unsigned short* src =
           G + (long)(half * 128 + row) * K + k0 + c_src * 8;
@@ -88,141 +84,150 @@ gemm8_kernel(const unsigned short* __restrict__ A,
           16, 0, 0);
     }
   };
-  auto stage_tile_half = [&](int k0, int buf, int h) {
-    // h: 0=A0, 1=B0, 2=A1, 3=B1 (A half first: consumed first)
-    if (h == 0) stage_half(Atile, k0, 0, buf, 0);
-    else if (h == 1) stage_half(Btile, k0, 1, buf, 0);
-    else if (h == 2) stage_half(Atile, k0, 0, buf, 1);
-    else stage_half(Btile, k0, 1, buf, 1);
-  };
 
-  // ---- fragment reads ------------------------------------------------
-  // A-frag (mi 0..7 within wave's 128 rows, kk 0..1):
-  //   row = mi*16 + (lane&15), chunk16 = kk*4 + (lane>>4)
+  // fragment read: (row16 = frag row-block within the half, kk)
   const int fr_row = lane & 15;
-  const int fr_c = lane >> 4;  // 0..3
+  const int fr_c = lane >> 4;
   auto read_frag = [&](const char* image, int row16, int kk) {
     return *reinterpret_cast<const bf16x8*>(
         image + g8_img_off(row16 * 16 + fr_row, kk * 4 + fr_c));
   };
 
-  f32x4 acc[8][4];
+  f32x4 acc[16][2];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < 16; ++i) {
+    acc[i][0] = {0.f, 0.f, 0.f, 0.f};
+    acc[i][1] = {0.f, 0.f, 0.f, 0.f};
+  }
 
   const int nt = K / G8_BK;
-  // prologue: tile 0 fully staged, synchronous
-  for (int h = 0; h < 4; ++h) stage_tile_half(0, 0, h);
+  // prologue: all of tile 0 + A-half0 of tile 1
+  stage_half(Atile, 0, 0, 0, 0);
+  stage_half(Atile, 0, 0, 0, 1);
+  stage_half(Btile, 0, 1, 0, 0);
+  stage_half(Btile, 0, 1, 0, 1);
+  if (nt > 1) stage_half(Atile, G8_BK, 0, 1, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
 
-  bf16x8 bfr[8];  // B frags: 4 nfrag x 2 kk (all cached per tile)
-  bf16x8 afr[8];  // A frags: 4 mi (current row-half) x 2 kk
+  bf16x8 afr[16];  // 8 mi (current row-half) x 2 kk
+  bf16x8 bfr[4];   // 2 ni x 2 kk (whole tile)
+  // wave's B rows within its half image: (wid&3)*32 + ni*16 + fr_row
+  const int bh = wid >> 2;        // B half index
+  const int brow0 = (wid & 3) * 32;
 
   for (int t = 0; t < nt; ++t) {
     const int buf = t & 1;
-    const char* Ai = (const char*)img(0, buf, wm);
-    const char* Bi = (const char*)img(1, buf, wn >> 1);
-    const int bn0 = (wn & 1) * 4;  // wave's 64 cols: nfrag 0..3 within
-                                   // its B-half (128 rows of B image)
-    const bool have_next = t + 1 < nt;
+    const char* Ai0 = (const char*)img(0, buf, 0);
+    const char* Ai1 = (const char*)img(0, buf, 1);
+    const char* Bi = (const char*)img(1, buf, bh);
     const int k1 = (t + 1) * G8_BK;
+    const int k2 = (t + 2) * G8_BK;
+    const bool n1 = t + 1 < nt, n2 = t + 2 < nt;
 
+    // ---- ph0: reads (A rh0 + B all), stage oth.A1 + oth.B0, MFMA ----
 #pragma unroll
-    for (int ph = 0; ph < 4; ++ph) {
-      // stage the next tile into the other buffer: 2 halves in ph0 and
-      // 2 in ph1, so by the vmcnt(0) at the END of ph3 every glds has
-      // had >= 2 phases (~1100 cycles of MFMA) to land.
-      if (have_next && ph == 0) {
-        stage_tile_half(k1, buf ^ 1, 0);
-        stage_tile_half(k1, buf ^ 1, 1);
-      } else if (have_next && ph == 1) {
-        stage_tile_half(k1, buf ^ 1, 2);
-        stage_tile_half(k1, buf ^ 1, 3);
-      }
-      if (ph == 0) {
-        // B frags for the whole tile + A frags of row-half 0
+    for (int ni = 0; ni < 2; ++ni)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+      for (int kk = 0; kk < 2; ++kk)
+        bfr[ni * 2 + kk] =
+            read_frag(Bi, (brow0 >> 4) + ni, kk);
 #pragma unroll
-          for (int kk = 0; kk < 2; ++kk)
-            bfr[ni * 2 + kk] = read_frag(Bi, bn0 + ni, kk);
+    for (int mi = 0; mi < 8; ++mi)
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-          for (int kk = 0; kk < 2; ++kk)
-            afr[mi * 2 + kk] = read_frag(Ai, mi, kk);
-      } else if (ph == 2) {
-        // A frags of row-half 1
-#pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-          for (int kk = 0; kk < 2; ++kk)
-            afr[mi * 2 + kk] = read_frag(Ai, 4 + mi, kk);
-      }
-      // quadrant (rh = ph>>1, ch = ph&1): 4 mi x 2 ni x 2 kk MFMAs
-      const int rh = ph >> 1;
-      const int ch = ph & 1;
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
-#pragma unroll
-          for (int kk = 0; kk < 2; ++kk)
-            acc[rh * 4 + mi][ch * 2 + ni] =
-                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    afr[mi * 2 + kk], bfr[(ch * 2 + ni) * 2 + kk],
-                    acc[rh * 4 + mi][ch * 2 + ni], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-      if (ph == 3) {
-        // every wave drains its own glds for tile t+1 BEFORE the
-        // barrier, so the next tile's cross-wave reads are safe.
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      // compiler memory fences around the raw barrier: keep LDS
-      // accesses on their side of it (s_barrier alone is not a fence)
-      asm volatile("" ::: "memory");
-      __builtin_amdgcn_s_barrier();
-      asm volatile("" ::: "memory");
+      for (int kk = 0; kk < 2; ++kk)
+        afr[mi * 2 + kk] = read_frag(Ai0, mi, kk);
+    if (n1) {
+      stage_half(Atile, k1, 0, buf ^ 1, 1);
+      stage_half(Btile, k1, 1, buf ^ 1, 0);
     }
-  }
-  // drain remaining staged-but-unused glds (none: last tile stages none)
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr[mi * 2 + kk], bfr[kk], acc[mi][0], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();   // mid barrier: A-rh0 reads done
+    asm volatile("" ::: "memory");
 
-  // ---- epilogue: 4 quadrant rounds through LDS ------------------------
-  // round q: each wave writes its quadrant rows q*32..q*32+31 (2 mfrag
-  // x 4 nfrag) as f32 [32][65] into its 8.3KiB slab, then streams it
-  // out coalesced with fused bias+act.
-  float* slab = reinterpret_cast<float*>(smem) + wid * 32 * 65;
-  const int grow0 = tm * G8_BM + wm * 128;
-  const int gcol0 = tn * G8_BN + wn * 64;
+    // ---- ph1: stage oth.B1 + cur.A0(t+2), MFMA (rh0, ch1) -----------
+    if (n1) stage_half(Btile, k1, 1, buf ^ 1, 1);
+    if (n2) stage_half(Atile, k2, 0, buf, 0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr[mi * 2 + kk], bfr[2 + kk], acc[mi][1], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph2: reads A rh1, MFMA (rh1, ch0) --------------------------
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        afr[mi * 2 + kk] = read_frag(Ai1, mi, kk);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        acc[8 + mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr[mi * 2 + kk], bfr[kk], acc[8 + mi][0], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph3: MFMA (rh1, ch1); counted wait; boundary barrier -------
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        acc[8 + mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr[mi * 2 + kk], bfr[2 + kk], acc[8 + mi][1], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    if (n2) {
+      // leave only A0(t+2) (the 2 newest glds) in flight
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();   // tile boundary
+    asm volatile("" ::: "memory");
+  }
+
+  // ---- epilogue: 4 rounds of 64 rows x 32 cols through padded LDS ----
+  float* slab = reinterpret_cast<float*>(smem) + wid * 64 * 33;
+  const int grow0 = tm * G8_BM;
+  const int gcol0 = tn * G8_BN + wid * 32;
   float bv = 0.f;
-  if (HAS_BIAS) bv = bf2f(bias[gcol0 + (lane & 63)]);
+  if (HAS_BIAS) bv = bf2f(bias[gcol0 + (lane & 31)]);
 #pragma unroll
   for (int qd = 0; qd < 4; ++qd) {
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi)
+    for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        const f32x4 v = acc[qd * 2 + mi][ni];
+      for (int ni = 0; ni < 2; ++ni) {
+        const f32x4 v = acc[qd * 4 + mi][ni];
         const int col = ni * 16 + (lane & 15);
         const int row0 = mi * 16 + (lane >> 4) * 4;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) slab[(row0 + r) * 65 + col] = v[r];
+        for (int r = 0; r < 4; ++r) slab[(row0 + r) * 33 + col] = v[r];
       }
     __syncthreads();
-    // stream 32 rows x 64 cols: lane covers one col, rows sequential
-    for (int r = 0; r < 32; ++r) {
-      float v = slab[r * 65 + lane];
+    for (int rr = 0; rr < 32; ++rr) {
+      const int r = rr * 2 + (lane >> 5);
+      float v = slab[r * 33 + (lane & 31)];
       if (HAS_BIAS) v += bv;
       if (ACT == G8_GELU_TANH) v = fast_gelu(v);
       else if (ACT == G8_GELU_ERF) v = gelu_erf(v);
       else if (ACT == G8_SILU) v = silu(v);
-      C[(long)(grow0 + qd * 32 + r) * N + gcol0 + lane] = f2bf(v);
+      C[(long)(grow0 + qd * 64 + r) * N + gcol0 + (lane & 31)] = f2bf(v);
     }
     __syncthreads();
   }
