@@ -87,7 +87,7 @@ class ChatAppConfig(BaseConfig):
 
 
 class ChatSession:
-    def __init__(self, config: ChatAppConfig):
+    def __init__(self, config: ChatAppConfig, batching: bool = False):
         self.config = config
         generator = get_generator(substitute_env(
             dict(config.generator_config)))
@@ -96,9 +96,23 @@ class ChatSession:
         self.rag = RagGenerator(generator=generator, retriever=retriever)
         self.template = ConversationPromptTemplate(config.system_prompt)
         self.last_retrieval: list[tuple[float, str]] = []
+        self.batcher = None
+        if batching:
+            from .engine.batcher import DynamicBatcher
 
-    def turn(self, user_text: str) -> str:
+            self.batcher = DynamicBatcher(generator)
+
+    def _generate(self, prompt: str) -> str:
+        if self.batcher is not None:
+            return self.batcher.generate_one(prompt)
+        return self.rag.generator.generate([prompt])[0]
+
+    def answer(self, history: list[tuple[str, str]], user_text: str
+               ) -> tuple[str, list[tuple[float, str]]]:
+        """Stateless turn: caller supplies history; returns (answer,
+        retrieval_dump). Thread-safe (used by the server per request)."""
         contexts = scores = None
+        retrieval: list[tuple[float, str]] = []
         if self.rag.retriever is not None:
             results, _ = self.rag.retriever.search(
                 [user_text], top_k=self.config.retrieval_top_k,
@@ -108,10 +122,16 @@ class ChatSession:
             texts = self.rag.retriever.get_texts(idxs)
             contexts = [texts]
             scores = [results.total_scores[0]]
-            self.last_retrieval = list(zip(results.total_scores[0], texts))
-        prompts = self.template.preprocess(user_text, contexts, scores)
-        responses = self.rag.generator.generate(prompts)
-        answer = self.template.postprocess(responses)[0]
+            retrieval = list(zip(results.total_scores[0], texts))
+        tmpl = ConversationPromptTemplate(self.config.system_prompt)
+        tmpl.history = list(history)
+        prompt = tmpl.preprocess(user_text, contexts, scores)[0]
+        answer = tmpl.postprocess([self._generate(prompt)])[0]
+        return answer, retrieval
+
+    def turn(self, user_text: str) -> str:
+        answer, retrieval = self.answer(self.template.history, user_text)
+        self.last_retrieval = retrieval
         self.template.history.append(("user", user_text))
         self.template.history.append(("assistant", answer))
         return answer

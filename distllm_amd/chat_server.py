@@ -39,7 +39,8 @@ def _get_session():
             config.retrieval_score_threshold = float(
                 os.environ["DISTLLM_SCORE_THRESHOLD"]
             )
-        _session = ChatSession(config)
+        batching = os.environ.get("DISTLLM_BATCHING", "1") != "0"
+        _session = ChatSession(config, batching=batching)
     return _session
 
 
@@ -74,7 +75,12 @@ async def chat_completions(req: ChatCompletionRequest):
     user_msgs = [m.content for m in req.messages if m.role == "user"]
     if not user_msgs:
         raise HTTPException(400, "no user message")
-    answer = await run_in_threadpool(session.turn, user_msgs[-1])
+    # stateless per request: client-supplied history, concurrent
+    # requests batch into one engine call (DynamicBatcher)
+    history = [(m.role, m.content) for m in req.messages[:-1]
+               if m.role in ("user", "assistant")]
+    answer, _retrieval = await run_in_threadpool(
+        session.answer, history, user_msgs[-1])
     rid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
     created = int(time.time())
     if req.stream:
