@@ -34,6 +34,8 @@ def main() -> None:
                     choices=["embed", "generate"],
                     help="embed = flagship seqs/s; generate = decode "
                          "tokens/s (Llama-class, paged KV)")
+    ap.add_argument("--pooler", type=str, default="mean",
+                    choices=["mean", "last_token"])
     ap.add_argument("--prompt-len", type=int, default=128)
     args = ap.parse_args()
 
@@ -75,7 +77,10 @@ def main() -> None:
         d_ids = ids_host.to(device, non_blocking=True)
         d_mask = mask_host.to(device, non_blocking=True)
         hidden = model(d_ids, d_mask)
-        pooled = ops.mean_pool(hidden, d_mask, normalize=True)
+        if args.pooler == "last_token":
+            pooled = ops.last_token_pool(hidden, d_mask, normalize=True)
+        else:
+            pooled = ops.mean_pool(hidden, d_mask, normalize=True)
         host.copy_(pooled, non_blocking=True)
 
     for _ in range(args.warmup):
@@ -103,9 +108,12 @@ def main() -> None:
     ms_per_step = elapsed / args.steps * 1000.0
     value = world * B * args.steps / elapsed
 
+    metric = ("seqs/sec embedded (PubMedBERT bs512)"
+              if args.model == "pubmedbert-base"
+              else f"seqs/sec embedded ({args.model} bs{B})")
     if rank == 0:
         print(json.dumps({
-            "metric": "seqs/sec embedded (PubMedBERT bs512)",
+            "metric": metric,
             "value": round(value, 2),
             "unit": "seqs/s",
             "n_gpus": world,

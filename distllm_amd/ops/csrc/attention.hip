@@ -32,7 +32,6 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
 #define KVBLK 64
 #define QWAVE 32
-#define QBLK 128  // 4 waves * 32
 #define LOG2E 1.4426950408889634f
 #define VROW 68   // padded V^T row length in bf16 elems (136 B)
 
@@ -52,11 +51,14 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
                 unsigned short* __restrict__ O, int Bsz, int SQ, int SKV,
                 int H, int HKV, float scale, long bstr_q, long tstr_q,
                 long bstr_k, long tstr_k) {
+  constexpr int NWAVE = 4;  // waves per workgroup (8 measured slower)
+  constexpr int NTHR = NWAVE * 64;
+  constexpr int QBLK = NWAVE * QWAVE;       // q rows per workgroup
   constexpr int NCH = D / 8;             // 16B chunks per row
   constexpr int DBLK = D / 32;           // 32-wide d blocks
   constexpr int KTILE_B = KVBLK * D * 2; // K tile bytes
   constexpr int VTILE_B = D * VROW * 2;  // V^T tile bytes
-  constexpr int VCH = KVBLK * NCH / 256; // V chunks per thread (2 or 4)
+  constexpr int VCH = KVBLK * NCH / NTHR;   // V chunks per thread
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -108,7 +110,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     constexpr int NPIECE = KVBLK / ROWS_PER;   // 8 / 16
     const int r_in = lane / NCH;
     const int c_dst = lane % NCH;
-    for (int p = wid; p < NPIECE; p += 4) {
+    for (int p = wid; p < NPIECE; p += NWAVE) {
       const int row = p * ROWS_PER + r_in;
       const int c_src = c_dst ^ (row & (NCH - 1));
       int krow = k0 + row;
@@ -122,12 +124,12 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     }
   };
   // per-thread V staging geometry, hoisted out of the tile loop
-  int v_k[VCH];                       // key index within tile
+  int v_k[VCH < 1 ? 1 : VCH];                       // key index within tile
   const unsigned short* v_src[VCH];   // &V[v_k][d0] (tile k0 = 0)
   int v_woff[VCH];                    // LDS elem offset d0*VROW + k
 #pragma unroll
   for (int c = 0; c < VCH; ++c) {
-    const int ci = threadIdx.x + c * 256;
+    const int ci = threadIdx.x + c * NTHR;
     v_k[c] = ci / NCH;
     const int d0 = (ci % NCH) * 8;
     v_src[c] = Vb + (long)v_k[c] * tstr_k + d0;
@@ -200,15 +202,18 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     const bool tile_masked = HAS_MASK || (k0 + KVBLK > SKV);
 
     if (wave_active) {
-      // ---- QK^T: two 32-key blocks --------------------------------
-      f32x16 accP[2];
-#pragma unroll
-      for (int kb = 0; kb < 2; ++kb)
-#pragma unroll
-        for (int i = 0; i < 16; ++i) accP[kb][i] = 0.f;
-      __builtin_amdgcn_s_setprio(1);
+      // ---- QK^T: two 32-key blocks, ONE live accumulator ----------
+      // (sequential kb keeps a single f32x16 accP: -16 VGPR, which is
+      // what lifts the D=128 variant from 1 to 2 waves/SIMD)
+      const float* mrow = reinterpret_cast<const float*>(smem + offM[cur]);
+      float p[32];
+      float pmax = -1e30f;
 #pragma unroll
       for (int kb = 0; kb < 2; ++kb) {
+        f32x16 accP;
+#pragma unroll
+        for (int i = 0; i < 16; ++i) accP[i] = 0.f;
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int s = 0; s < D / 16; ++s) {
           const int key = kb * 32 + (lane & 31);
@@ -216,27 +221,20 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
           const int swz = chunk ^ (key & (NCH - 1));
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
               smem + offK[cur] + key * (2 * D) + swz * 16);
-          accP[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              kf, qf[s], accP[kb], 0, 0, 0);
+          accP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kf, qf[s], accP, 0, 0, 0);
         }
-      }
-      __builtin_amdgcn_s_setprio(0);
-
-      // ---- softmax over 64 keys (32 per lane, halves combined) ----
-      const float* mrow = reinterpret_cast<const float*>(smem + offM[cur]);
-      float p[32];
-      float pmax = -1e30f;
-#pragma unroll
-      for (int kb = 0; kb < 2; ++kb)
+        __builtin_amdgcn_s_setprio(0);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int key_loc = kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * h;
-          float s2 = accP[kb][r] * scale2;
+          float s2 = accP[r] * scale2;
           if (tile_masked) s2 += mrow[key_loc];
           if (CAUSAL && (k0 + key_loc > qrow_abs)) s2 = -INFINITY;
           p[kb * 16 + r] = s2;
           pmax = fmaxf(pmax, s2);
         }
+      }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
       // defer-max (guide T13, THR=8 in log2 units): skip the O rescale
       // while the running max has not grown by more than THR — P is then
@@ -262,11 +260,13 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
       l = l * alpha + rs;
       m = mn;
 
-      // ---- P -> bf16 fragments (cvt + permlane32_swap) -------------
-      frag_u pa[4];
+      // ---- P -> bf16 fragments + PV, per kh (one pa live at a time:
+      // -12 VGPR vs materializing all four fragments) ----------------
+      const unsigned short* vimg =
+          reinterpret_cast<const unsigned short*>(smem + offV[cur]);
 #pragma unroll
-      for (int kb = 0; kb < 2; ++kb) {
-        const float* pp = p + kb * 16;
+      for (int kh = 0; kh < 4; ++kh) {
+        const float* pp = p + (kh >> 1) * 16 + (kh & 1) * 8;
         unsigned int a0 = ((unsigned int)f2bf(pp[0])) |
                           ((unsigned int)f2bf(pp[1]) << 16);
         unsigned int b0 = ((unsigned int)f2bf(pp[4])) |
@@ -275,47 +275,25 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
                           ((unsigned int)f2bf(pp[3]) << 16);
         unsigned int b1 = ((unsigned int)f2bf(pp[6])) |
                           ((unsigned int)f2bf(pp[7]) << 16);
-        unsigned int a2 = ((unsigned int)f2bf(pp[8])) |
-                          ((unsigned int)f2bf(pp[9]) << 16);
-        unsigned int b2 = ((unsigned int)f2bf(pp[12])) |
-                          ((unsigned int)f2bf(pp[13]) << 16);
-        unsigned int a3 = ((unsigned int)f2bf(pp[10])) |
-                          ((unsigned int)f2bf(pp[11]) << 16);
-        unsigned int b3 = ((unsigned int)f2bf(pp[14])) |
-                          ((unsigned int)f2bf(pp[15]) << 16);
         i32x2 s0 = __builtin_amdgcn_permlane32_swap((int)a0, (int)b0,
                                                     false, false);
         i32x2 s1 = __builtin_amdgcn_permlane32_swap((int)a1, (int)b1,
                                                     false, false);
-        i32x2 s2 = __builtin_amdgcn_permlane32_swap((int)a2, (int)b2,
-                                                    false, false);
-        i32x2 s3 = __builtin_amdgcn_permlane32_swap((int)a3, (int)b3,
-                                                    false, false);
-        pa[kb * 2].d[0] = (unsigned int)s0[0];
-        pa[kb * 2].d[1] = (unsigned int)s1[0];
-        pa[kb * 2].d[2] = (unsigned int)s0[1];
-        pa[kb * 2].d[3] = (unsigned int)s1[1];
-        pa[kb * 2 + 1].d[0] = (unsigned int)s2[0];
-        pa[kb * 2 + 1].d[1] = (unsigned int)s3[0];
-        pa[kb * 2 + 1].d[2] = (unsigned int)s2[1];
-        pa[kb * 2 + 1].d[3] = (unsigned int)s3[1];
-      }
-
-      // ---- PV: accO^T += V^T x P ------------------------------------
-      const unsigned short* vimg =
-          reinterpret_cast<const unsigned short*>(smem + offV[cur]);
-      __builtin_amdgcn_s_setprio(1);
+        frag_u pa;
+        pa.d[0] = (unsigned int)s0[0];
+        pa.d[1] = (unsigned int)s1[0];
+        pa.d[2] = (unsigned int)s0[1];
+        pa.d[3] = (unsigned int)s1[1];
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int d = 0; d < DBLK; ++d) {
-#pragma unroll
-        for (int kh = 0; kh < 4; ++kh) {
+        for (int d = 0; d < DBLK; ++d) {
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               vimg + (d * 32 + (lane & 31)) * VROW + kh * 16 + h * 8);
           accO[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              vf, pa[kh].v, accO[d], 0, 0, 0);
+              vf, pa.v, accO[d], 0, 0, 0);
         }
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
 
     if (have_next) {
@@ -374,19 +352,21 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
                                 float scale, long bstr_q, long tstr_q,
                                 long bstr_k, long tstr_k,
                                 hipStream_t stream) {
-  const int nqb = (SQ + QBLK - 1) / QBLK;
+  const int nwave = 4;
+  const int qblk = nwave * 32;
+  const int nqb = (SQ + qblk - 1) / qblk;
   const int grid = B * H * nqb;
   const int ktile = KVBLK * D * 2, vtile = D * VROW * 2;
   const int stage_b = 2 * ktile + 2 * vtile + 2 * KVBLK * 4;
-  const int epi_b = 4 * QWAVE * (D + 1) * 4;
+  const int epi_b = nwave * 32 * (D + 1) * 4;
   const int lds = stage_b > epi_b ? stage_b : epi_b;
 
 #define ATTN_DISPATCH(DD, CC, MM)                                              \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM>), dim3(grid), dim3(256),     \
-                     lds, stream, (const unsigned short*)Q,                    \
-                     (const unsigned short*)K, (const unsigned short*)V, mask, \
-                     (unsigned short*)O, B, SQ, SKV, H, HKV, scale, bstr_q,    \
-                     tstr_q, bstr_k, tstr_k)
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM>), dim3(grid),                \
+                     dim3(nwave * 64), lds, stream,                            \
+                     (const unsigned short*)Q, (const unsigned short*)K,       \
+                     (const unsigned short*)V, mask, (unsigned short*)O, B,    \
+                     SQ, SKV, H, HKV, scale, bstr_q, tstr_q, bstr_k, tstr_k)
   const bool hm = mask != nullptr;
   if (D == 64) {
     if (causal && hm) ATTN_DISPATCH(64, true, true);

@@ -78,3 +78,40 @@ def test_vllm_alias_maps_to_native():
     from distllm_amd.generate.generators import STRATEGIES
 
     assert STRATEGIES["vllm"][1].__name__ == "NativeGenerator"
+
+
+def test_dynamic_batcher_groups_requests():
+    from concurrent.futures import ThreadPoolExecutor
+
+    from distllm_amd.engine.batcher import DynamicBatcher
+
+    calls = []
+
+    class FakeGen:
+        def generate(self, prompts):
+            calls.append(len(prompts))
+            return [p.upper() for p in prompts]
+
+    b = DynamicBatcher(FakeGen(), max_batch=8, window_ms=50)
+    with ThreadPoolExecutor(max_workers=6) as ex:
+        futs = [ex.submit(b.generate_one, f"p{i}") for i in range(6)]
+        outs = [f.result(timeout=30) for f in futs]
+    assert sorted(outs) == [f"P{i}" for i in range(6)]
+    # concurrent requests were grouped (fewer calls than requests)
+    assert sum(calls) == 6 and len(calls) < 6
+    b.close()
+
+
+def test_dynamic_batcher_error_propagation():
+    from distllm_amd.engine.batcher import DynamicBatcher
+
+    class BadGen:
+        def generate(self, prompts):
+            raise RuntimeError("engine down")
+
+    b = DynamicBatcher(BadGen(), window_ms=1)
+    import pytest as _pytest
+
+    with _pytest.raises(RuntimeError):
+        b.generate_one("x", timeout=10)
+    b.close()
