@@ -43,6 +43,7 @@ class EmbeddingConfig(BaseConfig):
     embedder_config: dict[str, Any]
     writer_config: dict[str, Any]
     compute_config: dict[str, Any] = {"name": "local"}
+    retries: int = 1  # per-file retries (reference parsl.py:85,130,197)
 
 
 def embedding_worker(
@@ -90,15 +91,28 @@ def main(config: EmbeddingConfig) -> None:
 
     emb_dir = config.output_dir / "embeddings"
     emb_dir.mkdir(exist_ok=True, parents=True)
+    failed: list[str] = []
     for path in my_files:
-        with Timer("finished-embedding", str(path), f"rank{rank}"):
-            embedding_worker(
-                path, emb_dir,
-                config.dataset_config, config.encoder_config,
-                config.pooler_config, config.embedder_config,
-                config.writer_config,
-            )
+        # per-file retry replaces Parsl task retries: a failed file only
+        # loses itself (share-nothing), and the run reports it at the end
+        for attempt in range(config.retries + 1):
+            try:
+                with Timer("finished-embedding", str(path), f"rank{rank}"):
+                    embedding_worker(
+                        path, emb_dir,
+                        config.dataset_config, config.encoder_config,
+                        config.pooler_config, config.embedder_config,
+                        config.writer_config,
+                    )
+                break
+            except Exception as e:
+                print(f"[rank {rank}] {path} attempt {attempt + 1} "
+                      f"failed: {e!r}")
+                if attempt == config.retries:
+                    failed.append(str(path))
     barrier()
+    if failed:
+        print(f"[rank {rank}] FAILED files: {failed}")
     if rank == 0:
         print(f"[done] {len(files)} files -> {emb_dir}")
 
