@@ -27,6 +27,7 @@
 // [B, SKV, HKV, D] (shared strides), O contiguous BSHD; mask f32
 // [B, SKV] additive (0 / -inf) or nullptr.
 #include "common.h"
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
@@ -42,7 +43,7 @@ union frag_u {
   unsigned int d[4];
 };
 
-template <int D, bool CAUSAL, bool HAS_MASK>
+template <int D, bool CAUSAL, bool HAS_MASK, bool VDIRECT>
 __global__ void __launch_bounds__(256, 1)
 attn_fwd_kernel(const unsigned short* __restrict__ Q,
                 const unsigned short* __restrict__ Kt,
@@ -179,8 +180,10 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   // prologue: stage tile 0 synchronously
   s16x8 vreg[VCH];
   stage_k(0, 0);
-  v_issue(0, vreg);
-  v_write(0, vreg);
+  if (!VDIRECT) {
+    v_issue(0, vreg);
+    v_write(0, vreg);
+  }
   stage_m(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -191,7 +194,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     const bool have_next = t + 1 < ntile;
     if (have_next) {
       stage_k(k0 + KVBLK, cur ^ 1);   // async glds into the other buffer
-      v_issue(k0 + KVBLK, vreg);      // V -> regs, no wait (T14)
+      if (!VDIRECT) v_issue(k0 + KVBLK, vreg);  // V -> regs (T14)
     }
 
     const bool wave_active =
@@ -287,8 +290,21 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int d = 0; d < DBLK; ++d) {
-          bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-              vimg + (d * 32 + (lane & 31)) * VROW + kh * 16 + h * 8);
+          bf16x8 vf;
+          if (VDIRECT) {
+            // strided-but-coalesced global reads (per i: 64 lanes hit
+            // 2 key rows of consecutive d) served from L1/L2
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+              int key = k0 + kh * 16 + h * 8 + i;
+              if (key >= SKV) key = SKV - 1;
+              vf[i] = *reinterpret_cast<const __bf16*>(
+                  Vb + (long)key * tstr_k + d * 32 + (lane & 31));
+            }
+          } else {
+            vf = *reinterpret_cast<const bf16x8*>(
+                vimg + (d * 32 + (lane & 31)) * VROW + kh * 16 + h * 8);
+          }
           accO[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               vf, pa.v, accO[d], 0, 0, 0);
         }
@@ -298,7 +314,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
 
     if (have_next) {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // V regs + glds
-      v_write(cur ^ 1, vreg);
+      if (!VDIRECT) v_write(cur ^ 1, vreg);
       stage_m(k0 + KVBLK, cur ^ 1);
     }
     __syncthreads();  // ONE barrier per tile
@@ -361,12 +377,28 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
   const int epi_b = nwave * 32 * (D + 1) * 4;
   const int lds = stage_b > epi_b ? stage_b : epi_b;
 
+  static int vdirect = -1;
+  if (vdirect < 0) {
+    const char* e = getenv("DISTLLM_ATTN_VDIRECT");
+    vdirect = (e && e[0] == '1') ? 1 : 0;
+  }
 #define ATTN_DISPATCH(DD, CC, MM)                                              \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM>), dim3(grid),                \
-                     dim3(nwave * 64), lds, stream,                            \
-                     (const unsigned short*)Q, (const unsigned short*)K,       \
-                     (const unsigned short*)V, mask, (unsigned short*)O, B,    \
-                     SQ, SKV, H, HKV, scale, bstr_q, tstr_q, bstr_k, tstr_k)
+  do {                                                                         \
+    if (vdirect)                                                               \
+      hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM, true>), dim3(grid),      \
+                         dim3(nwave * 64), lds, stream,                        \
+                         (const unsigned short*)Q, (const unsigned short*)K,   \
+                         (const unsigned short*)V, mask, (unsigned short*)O,   \
+                         B, SQ, SKV, H, HKV, scale, bstr_q, tstr_q, bstr_k,    \
+                         tstr_k);                                              \
+    else                                                                       \
+      hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM, false>), dim3(grid),     \
+                         dim3(nwave * 64), lds, stream,                        \
+                         (const unsigned short*)Q, (const unsigned short*)K,   \
+                         (const unsigned short*)V, mask, (unsigned short*)O,   \
+                         B, SQ, SKV, H, HKV, scale, bstr_q, tstr_q, bstr_k,    \
+                         tstr_k);                                              \
+  } while (0)
   const bool hm = mask != nullptr;
   if (D == 64) {
     if (causal && hm) ATTN_DISPATCH(64, true, true);
