@@ -377,14 +377,18 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
   const int epi_b = nwave * 32 * (D + 1) * 4;
   const int lds = stage_b > epi_b ? stage_b : epi_b;
 
+  // measured on MI355X: LDS-staged V wins at D=64 (330 vs 302 TF),
+  // direct-global V wins at D=128 (176 vs 105 TF — drops the VGPR
+  // pressure that pinned the kernel at 1 wave/SIMD). Env-overridable.
   static int vdirect = -1;
   if (vdirect < 0) {
     const char* e = getenv("DISTLLM_ATTN_VDIRECT");
-    vdirect = (e && e[0] == '1') ? 1 : 0;
+    vdirect = e ? (e[0] == '1' ? 1 : 0) : -1;
   }
+  const int vd = (vdirect >= 0) ? vdirect : (D == 128 ? 1 : 0);
 #define ATTN_DISPATCH(DD, CC, MM)                                              \
   do {                                                                         \
-    if (vdirect)                                                               \
+    if (vd)                                                               \
       hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM, true>), dim3(grid),      \
                          dim3(nwave * 64), lds, stream,                        \
                          (const unsigned short*)Q, (const unsigned short*)K,   \
