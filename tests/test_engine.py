@@ -115,3 +115,19 @@ def test_dynamic_batcher_error_propagation():
     with _pytest.raises(RuntimeError):
         b.generate_one("x", timeout=10)
     b.close()
+
+
+def test_engine_truncates_long_prompts():
+    import torch
+
+    from distllm_amd.engine import LLMEngine, SamplingParams
+    from distllm_amd.models import build_model
+
+    torch.manual_seed(3)
+    m = build_model("llama-tiny", dtype=torch.float32)
+    engine = LLMEngine(m, max_batch_size=2, max_model_len=32,
+                       num_blocks=16)
+    params = SamplingParams(temperature=0.0, min_p=0.0, max_tokens=2)
+    long_prompt = list(range(5, 105))  # 100 tokens >> max_model_len
+    outs = engine.generate([long_prompt], params)
+    assert len(outs[0]) >= 1  # generated within the truncated window
