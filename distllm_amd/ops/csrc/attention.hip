@@ -84,12 +84,10 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int qrow_abs = q0w + qcol;
   const int qload = qrow_abs < SQ ? qrow_abs : (SQ > 0 ? SQ - 1 : 0);
 
-  // K is TRIPLE buffered: its glds stay in flight across the raw
-  // barrier (counted vmcnt), two tiles ahead of consumption.
-  const int offK[3] = {0, KTILE_B, 2 * KTILE_B};
-  const int offV[2] = {3 * KTILE_B, 3 * KTILE_B + VTILE_B};
-  const int offM[2] = {3 * KTILE_B + 2 * VTILE_B,
-                       3 * KTILE_B + 2 * VTILE_B + KVBLK * 4};
+  const int offK[2] = {0, KTILE_B};
+  const int offV[2] = {2 * KTILE_B, 2 * KTILE_B + VTILE_B};
+  const int offM[2] = {2 * KTILE_B + 2 * VTILE_B,
+                       2 * KTILE_B + 2 * VTILE_B + KVBLK * 4};
 
   const float scale2 = scale * LOG2E;
 
@@ -179,10 +177,9 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     ntile = min(ntile, qmax_blk / KVBLK + 1);
   }
 
-  // prologue: K tiles 0 and 1, V/mask tile 0, full drain once
+  // prologue: stage tile 0 synchronously
   s16x8 vreg[VCH];
   stage_k(0, 0);
-  if (ntile > 1) stage_k(KVBLK, 1);
   if (!VDIRECT) {
     v_issue(0, vreg);
     v_write(0, vreg);
@@ -191,19 +188,14 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
-  // glds instructions per wave per K tile (= the counted vmcnt target)
-  constexpr int KP = (KVBLK / (1024 / (2 * D))) / NWAVE;
-
+  int cur = 0;
   for (int t = 0; t < ntile; ++t) {
     const int k0 = t * KVBLK;
-    const int kcur = t % 3;
-    const int cur = t & 1;               // V/mask buffer
     const bool have_next = t + 1 < ntile;
-    const bool have_next2 = t + 2 < ntile;
-    // V loads first, K glds second: the tile-end counted wait drains V
-    // (and the previous K) while leaving this K prefetch in flight.
-    if (have_next && !VDIRECT) v_issue(k0 + KVBLK, vreg);
-    if (have_next2) stage_k(k0 + 2 * KVBLK, (t + 2) % 3);
+    if (have_next) {
+      stage_k(k0 + KVBLK, cur ^ 1);   // async glds into the other buffer
+      if (!VDIRECT) v_issue(k0 + KVBLK, vreg);  // V -> regs (T14)
+    }
 
     const bool wave_active =
         q0w < SQ && (!CAUSAL || k0 <= q0w + QWAVE - 1);
@@ -231,7 +223,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
           const int chunk = (s * 16 + h * 8) / 8;
           const int swz = chunk ^ (key & (NCH - 1));
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              smem + offK[kcur] + key * (2 * D) + swz * 16);
+              smem + offK[cur] + key * (2 * D) + swz * 16);
           accP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               kf, qf[s], accP, 0, 0, 0);
         }
@@ -320,25 +312,13 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
       }
     }
 
-    // counted tile-end wait: everything older than the K(t+2) prefetch
-    // (i.e. this tile's V loads and the K(t+1) glds) has landed.
-    if (have_next2) {
-      if constexpr (KP == 2)
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-      else
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
     if (have_next) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // V regs + glds
       if (!VDIRECT) v_write(cur ^ 1, vreg);
       stage_m(k0 + KVBLK, cur ^ 1);
     }
-    // raw barrier: drain only the LDS queue (ds_writes above), keep the
-    // K glds in flight across the barrier.
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    asm volatile("" ::: "memory");
+    __syncthreads();  // ONE barrier per tile
+    cur ^= 1;
   }
 
   // ---- epilogue: O^T regs -> padded LDS -> coalesced bf16 stores ------
@@ -393,7 +373,7 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
   const int nqb = (SQ + qblk - 1) / qblk;
   const int grid = B * H * nqb;
   const int ktile = KVBLK * D * 2, vtile = D * VROW * 2;
-  const int stage_b = 3 * ktile + 2 * vtile + 2 * KVBLK * 4;
+  const int stage_b = 2 * ktile + 2 * vtile + 2 * KVBLK * 4;
   const int epi_b = nwave * 32 * (D + 1) * 4;
   const int lds = stage_b > epi_b ? stage_b : epi_b;
 
