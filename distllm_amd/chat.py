@@ -16,7 +16,6 @@ import time
 from pathlib import Path
 from typing import Any
 
-from pydantic import Field
 
 from .generate import get_generator
 from .rag.response_synthesizer import RagGenerator

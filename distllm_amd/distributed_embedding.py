@@ -29,7 +29,7 @@ from .embed import (
     get_writer,
     unique_shard_dir,
 )
-from .parallel import barrier, get_rank, init_distributed, shard_list
+from .parallel import barrier, init_distributed, shard_list
 from .utils import BaseConfig, Timer
 
 

@@ -10,7 +10,6 @@ MI355X; fp32 is available for CPU testing).
 
 from __future__ import annotations
 
-from pathlib import Path
 from typing import Any, Literal, Union
 
 import torch

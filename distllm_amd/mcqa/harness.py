@@ -29,8 +29,6 @@ from concurrent.futures import ThreadPoolExecutor, as_completed
 from pathlib import Path
 from typing import Any
 
-from pydantic import Field
-
 from ..utils import BaseConfig, retry
 
 # ---------------------------------------------------------------------------

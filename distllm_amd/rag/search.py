@@ -28,8 +28,8 @@ import torch
 from pydantic import Field
 
 from .. import ops
-from ..embed.encoders import EncoderConfigs, get_encoder
-from ..embed.poolers import PoolerConfigs, get_pooler
+from ..embed.encoders import get_encoder
+from ..embed.poolers import get_pooler
 from ..utils import BaseConfig, batch_data
 
 
