@@ -165,11 +165,21 @@ def _autotune_gemm(x2, weight, bias, act):
     return "gemm8" if t_hip < t_blas else "blaslt"
 
 
+def _decode_splitk(M, N, K):
+    """SPLITK so (N/64)*SPLITK lands near the CU count (256)."""
+    ncb = N // 64
+    for sk in (4, 2, 1):
+        if ncb * sk <= 512 and K % (64 * sk) == 0:
+            return sk
+    return 1
+
+
 def linear(x, weight, bias=None, act=ACT_NONE):
     """Linear layer entry used by models. GPU path: autotuned choice
-    between hipBLASLt (+fused activation kernel) and the hand-written
-    deep-pipelined MFMA GEMM with fused bias+activation epilogue
-    (gemm8.hip). Override with DISTLLM_GEMM=blaslt|gemm8."""
+    between hipBLASLt (+fused activation kernel), the hand-written
+    deep-pipelined MFMA GEMM (gemm8.hip), and the skinny-M split-K
+    decode GEMM (decode_gemm.hip) for M <= 256. Override with
+    DISTLLM_GEMM=blaslt|gemm8."""
     import os
 
     orig_shape = x.shape
@@ -179,6 +189,34 @@ def linear(x, weight, bias=None, act=ACT_NONE):
     if x.is_cuda:
         _require_hip()
         mode = os.environ.get("DISTLLM_GEMM", "auto")
+        if (mode != "blaslt" and act == ACT_NONE and M <= 256
+                and N % 64 == 0 and K % 64 == 0 and x2.is_contiguous()):
+            key = ("dec", M, N, K, bias is not None)
+            choice = _GEMM_CHOICE.get(key)
+            sk = _decode_splitk(M, N, K)
+            if choice is None:
+                import time as _t
+
+                def timeit(fn):
+                    fn()
+                    torch.cuda.synchronize()
+                    t0 = _t.perf_counter()
+                    for _ in range(10):
+                        fn()
+                    torch.cuda.synchronize()
+                    return _t.perf_counter() - t0
+
+                tb = timeit(lambda: torch.nn.functional.linear(
+                    x2, weight, bias))
+                td = timeit(lambda: _hip_ops.decode_gemm(x2, weight, bias,
+                                                         sk))
+                choice = "decode" if td < tb else "blaslt"
+                _GEMM_CHOICE[key] = choice
+            if choice == "decode":
+                out = _hip_ops.decode_gemm(x2, weight, bias, sk)
+                return out.reshape(*orig_shape[:-1], N)
+            out = torch.nn.functional.linear(x2, weight, bias)
+            return out.reshape(*orig_shape[:-1], N)
         shapes_ok = (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
                      and x2.is_contiguous())
         if shapes_ok and mode != "blaslt":

@@ -35,6 +35,8 @@ void launch_gemm_bf16(const void*, const void*, const void*, void*, int, int,
                       int, int, hipStream_t);
 void launch_gemm8_bf16(const void*, const void*, const void*, void*, int, int,
                        int, int, hipStream_t);
+void launch_decode_gemm(const void*, const void*, const void*, void*, void*,
+                        int, int, int, int, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, const float*,
                      void*, int, int, int, int, int, int, bool, float, long,
                      long, long, long, hipStream_t);
@@ -171,6 +173,31 @@ torch::Tensor gemm8_bf16(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+torch::Tensor decode_gemm(torch::Tensor X, torch::Tensor W,
+                          c10::optional<torch::Tensor> bias,
+                          int64_t splitk) {
+  CHECK_DEV(X); CHECK_CONT(X); CHECK_BF16(X);
+  CHECK_DEV(W); CHECK_CONT(W); CHECK_BF16(W);
+  const int M = X.size(0), K = X.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == K, "K mismatch");
+  TORCH_CHECK(M <= 256, "decode_gemm is for M <= 256");
+  TORCH_CHECK(N % 64 == 0, "N % 64");
+  TORCH_CHECK(K % (64 * splitk) == 0, "K % (64*splitk)");
+  const int MP = M <= 128 ? 128 : 256;
+  auto out = torch::empty({M, N}, X.options());
+  torch::Tensor slabs;
+  void* slabp = nullptr;
+  if (splitk > 1) {
+    slabs = torch::empty({splitk, (long)MP, (long)N},
+                         X.options().dtype(at::kFloat));
+    slabp = slabs.data_ptr();
+  }
+  launch_decode_gemm(X.data_ptr(), W.data_ptr(),
+                     bias ? bias->data_ptr() : nullptr, out.data_ptr(),
+                     slabp, M, N, K, (int)splitk, cur_stream());
+  return out;
+}
+
 // ---- attention -------------------------------------------------------------
 torch::Tensor attn_fwd(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
                        c10::optional<torch::Tensor> mask, bool causal,
@@ -300,6 +327,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adjacent_cosine", &adjacent_cosine);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("gemm8_bf16", &gemm8_bf16);
+  m.def("decode_gemm", &decode_gemm);
   m.def("attn_fwd", &attn_fwd);
   m.def("rope_apply", &rope_apply);
   m.def("topk_chunked", &topk_chunked);

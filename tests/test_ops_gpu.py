@@ -242,3 +242,17 @@ def test_cache_put(device):
     ops.cache_put(kn, vn, kcache, vcache, bt, slot)
     assert torch.equal(kcache[1, :, 1].cpu(), kn[0].cpu())  # b0 pos17 blk1
     assert torch.equal(vcache[max_blocks, :, 3].cpu(), vn[1].cpu())
+
+
+@pytest.mark.parametrize("M,N,K,sk", [(128, 4096, 4096, 4),
+                                      (128, 14336, 4096, 1),
+                                      (96, 4096, 14336, 4),
+                                      (256, 6144, 4096, 2)])
+def test_decode_gemm(device, M, N, K, sk):
+    torch.manual_seed(M + N)
+    x = torch.randn(M, K, device=device, dtype=torch.bfloat16) * 0.2
+    w = torch.randn(N, K, device=device, dtype=torch.bfloat16) * 0.2
+    b = torch.randn(N, device=device, dtype=torch.bfloat16)
+    out = ops._hip_ops.decode_gemm(x, w, b, sk)
+    ref = torch_ref.gemm(x, w, b)
+    _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
