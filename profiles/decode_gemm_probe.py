@@ -32,15 +32,19 @@ def main():
         ("down", 4096, 14336),
         ("lm_head", 128256, 4096),
     ]
+    import distllm_amd.ops as ops
+
     for name, N, K in shapes:
         x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
         w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
         t = bench(lambda: torch.nn.functional.linear(x, w))
+        sk = ops._decode_splitk(M, N, K)
+        td = bench(lambda: ops._hip_ops.decode_gemm(x, w, None, sk))
         wbytes = N * K * 2
         roof = wbytes / 6.3e12
-        print(f"M{M} {name:<11} N{N} K{K}: {t * 1e6:8.1f} us  "
-              f"roofline {roof * 1e6:7.1f} us  eff {100 * roof / t:5.1f}%  "
-              f"BW {wbytes / t / 1e12:.2f} TB/s")
+        print(f"M{M} {name:<11} N{N} K{K}: blaslt {t * 1e6:7.1f}us "
+              f"({100 * roof / t:4.1f}%) | decode sk{sk} {td * 1e6:7.1f}us "
+              f"({100 * roof / td:4.1f}%)")
 
 
 if __name__ == "__main__":
