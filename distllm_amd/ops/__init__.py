@@ -166,9 +166,10 @@ def _autotune_gemm(x2, weight, bias, act):
 
 
 def _decode_splitk(M, N, K):
-    """SPLITK so (N/64)*SPLITK lands near the CU count (256)."""
+    """SPLITK so (N/64)*SPLITK lands near the CU count (256-512):
+    narrow projections (k/v heads) need deep K splits to fill the chip."""
     ncb = N // 64
-    for sk in (4, 2, 1):
+    for sk in (16, 8, 4, 2, 1):
         if ncb * sk <= 512 and K % (64 * sk) == 0:
             return sk
     return 1
