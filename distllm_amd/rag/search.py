@@ -67,6 +67,8 @@ class HipIndexConfig(BaseConfig):
     search_algorithm: str = "exact"
     rescore_multiplier: int = 2
     num_quantization_workers: int = 1
+    nlist: int = 0    # ivf: clusters (0 -> ~sqrt(N))
+    nprobe: int = 8   # ivf: clusters searched per query
 
 
 class HipFlatIndex:
@@ -80,14 +82,20 @@ class HipFlatIndex:
                  rescore_multiplier: int = 2,
                  num_quantization_workers: int = 1,
                  device: str | None = None,
-                 score_chunk_rows: int = 1 << 21):
+                 score_chunk_rows: int = 1 << 21,
+                 nlist: int = 0, nprobe: int = 8):
         if precision not in ("float32", "ubinary"):
             raise ValueError(f"invalid precision {precision}")
-        if search_algorithm not in ("exact", "hnsw"):
+        if search_algorithm not in ("exact", "hnsw", "ivf"):
             raise ValueError(f"invalid search_algorithm {search_algorithm}")
         if search_algorithm == "hnsw":
             print("[hip_index] 'hnsw' requested: exact GPU flat search is "
-                  "used instead (covers reference defaults)")
+                  "used instead (covers reference defaults; 'ivf' gives "
+                  "an approximate option)")
+            search_algorithm = "exact"
+        self.search_algorithm = search_algorithm
+        self.nlist = nlist
+        self.nprobe = nprobe
         self.precision = precision
         self.rescore_multiplier = rescore_multiplier
         self.faiss_index_path = Path(faiss_index_path)
@@ -150,6 +158,57 @@ class HipFlatIndex:
             c64 = c64.reshape(self.n, W).T.copy()  # word-major [W, N]
             self.codes_t = torch.from_numpy(c64.view(np.int64)).to(dev)
             self.code_words = W
+        if self.search_algorithm == "ivf":
+            self._build_ivf()
+
+    def _build_ivf(self) -> None:
+        """Simple GPU k-means (inner-product on normalized centroids);
+        corpus reordered cluster-major for contiguous candidate gathers.
+        Replaces the reference's faiss HNSW approximate option
+        (rag/search.py:237,245) with an IVF-flat design that maps to
+        GEMM + gather on this hardware."""
+        C = self.nlist or max(int(self.n ** 0.5), 1)
+        C = min(C, self.n)
+        g = torch.Generator().manual_seed(0)
+        init = torch.randperm(self.n, generator=g)[:C]
+        cent = self.corpus[init.to(self.corpus.device)].clone()
+        for _ in range(10):
+            assign = (self.corpus @ cent.t()).argmax(1)
+            for c in range(C):
+                m = assign == c
+                if m.any():
+                    cent[c] = self.corpus[m].mean(0)
+        assign = (self.corpus @ cent.t()).argmax(1)
+        order = torch.argsort(assign, stable=True)
+        counts = torch.bincount(assign, minlength=C)
+        offs = torch.zeros(C + 1, dtype=torch.long)
+        offs[1:] = counts.cumsum(0).cpu()
+        self.ivf_centroids = cent
+        self.ivf_order = order            # corpus row ids, cluster-major
+        self.ivf_offsets = offs           # [C+1] boundaries into order
+        self.ivf_nlist = C
+
+    def _topk_ivf(self, q: torch.Tensor, k: int):
+        nprobe = min(self.nprobe, self.ivf_nlist)
+        cscores = q @ self.ivf_centroids.t()
+        _, probes = torch.topk(cscores, nprobe, dim=1)   # [Q, nprobe]
+        vs, is_ = [], []
+        for qi in range(q.shape[0]):
+            cand = torch.cat([
+                self.ivf_order[self.ivf_offsets[c]:self.ivf_offsets[c + 1]]
+                for c in probes[qi].tolist()
+            ])
+            scores = (self.corpus[cand] @ q[qi]).unsqueeze(0)
+            kk = min(k, cand.numel())
+            v, i = ops.topk(scores.contiguous(), kk)
+            idx = cand[i[0].long()].int()
+            if kk < k:
+                v = torch.nn.functional.pad(v, (0, k - kk),
+                                            value=float("-inf"))
+                idx = torch.nn.functional.pad(idx, (0, k - kk), value=-1)
+            vs.append(v[0])
+            is_.append(idx)
+        return torch.stack(vs), torch.stack(is_)
 
     # ---- search ---------------------------------------------------------
     def transform(self, embeddings: np.ndarray) -> np.ndarray:
@@ -203,6 +262,8 @@ class HipFlatIndex:
         ).to(self.corpus.device)
         if self.precision == "ubinary":
             v, i = self._topk_ubinary(q, top_k)
+        elif self.search_algorithm == "ivf":
+            v, i = self._topk_ivf(q, top_k)
         else:
             v, i = self._topk_exact(q, top_k)
         v = v.cpu().tolist()

@@ -116,3 +116,35 @@ def test_retriever_and_rag_generator(tmp_path):
                            {"name": "question_answer"}),
                        retrieval_top_k=2)
     assert len(out) == 1 and isinstance(out[0], str)
+
+
+def test_ivf_search(tmp_path):
+    from datasets import Dataset as HFDataset
+
+    rng = np.random.default_rng(9)
+    # clustered corpus: 20 centers x 40 points
+    centers = rng.standard_normal((20, 24)).astype(np.float32) * 3
+    emb = np.concatenate([
+        c + 0.1 * rng.standard_normal((40, 24)).astype(np.float32)
+        for c in centers
+    ])
+    emb = normalize_L2(emb)
+    d = tmp_path / "c"
+    HFDataset.from_dict({
+        "text": [f"t{i}" for i in range(len(emb))],
+        "embeddings": [e for e in emb],
+    }).save_to_disk(str(d))
+    idx = HipFlatIndex(d, tmp_path / "i.npz", search_algorithm="ivf",
+                       nlist=20, nprobe=4)
+    q = emb[[5, 333, 777]]
+    res = idx.search(q, top_k=5)
+    for qi, want in zip(range(3), [5, 333, 777]):
+        assert res.total_indices[qi][0] == want
+    # recall vs exact on clustered data should be high
+    exact = HipFlatIndex(d, tmp_path / "e.npz")
+    ref = exact.search(q, top_k=5)
+    overlap = sum(
+        len(set(a) & set(b))
+        for a, b in zip(res.total_indices, ref.total_indices)
+    ) / 15.0
+    assert overlap >= 0.8, overlap
