@@ -51,7 +51,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
                 const float* __restrict__ maskp,
                 unsigned short* __restrict__ O, int Bsz, int SQ, int SKV,
                 int H, int HKV, float scale, long bstr_q, long tstr_q,
-                long bstr_k, long tstr_k, int qbpb) {
+                long bstr_k, long tstr_k) {
   constexpr int NWAVE = 4;  // waves per workgroup (8 measured slower)
   constexpr int NTHR = NWAVE * 64;
   constexpr int QBLK = NWAVE * QWAVE;       // q rows per workgroup
@@ -67,26 +67,12 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int qcol = lane & 31;
 
   const int nqb = (SQ + QBLK - 1) / QBLK;
-  // One workgroup iterates qbpb q-blocks of ONE (b, head): the K/V
-  // stream repeats with high temporal locality, so re-reads come from
-  // L2 instead of HBM (attention at these shapes is otherwise bound by
-  // the nqb-fold K/V re-read — measured to match the HBM roofline).
-  // T1 bijective XCD swizzle keeps consecutive ids on one XCD.
-  const int ngrp = (nqb + qbpb - 1) / qbpb;
-  int bid = blockIdx.x;
-  {
-    const int nwg = Bsz * H * ngrp;
-    const int q = nwg / 8, r = nwg % 8;
-    const int xcd = bid % 8, j = bid / 8;
-    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + j;
-  }
-  const int grp = bid % ngrp;
-  const int bh = bid / ngrp;
+  const int bid = blockIdx.x;
+  const int qb = bid % nqb;
+  const int bh = bid / nqb;
   const int head = bh % H;
   const int b = bh / H;
   const int kvhead = head / (H / HKV);
-  const int qb_lo = grp * qbpb;
-  const int qb_hi = min(qb_lo + qbpb, nqb);
 
   const long ostride = (long)H * D;
   const unsigned short* Qb = Q + (long)b * bstr_q + (long)head * D;
@@ -94,12 +80,30 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const unsigned short* Vb = Vt + (long)b * bstr_k + (long)kvhead * D;
   unsigned short* Ob = O + ((long)b * SQ) * ostride + (long)head * D;
 
+  const int q0w = qb * QBLK + wid * QWAVE;
+  const int qrow_abs = q0w + qcol;
+  const int qload = qrow_abs < SQ ? qrow_abs : (SQ > 0 ? SQ - 1 : 0);
+
   const int offK[2] = {0, KTILE_B};
   const int offV[2] = {2 * KTILE_B, 2 * KTILE_B + VTILE_B};
   const int offM[2] = {2 * KTILE_B + 2 * VTILE_B,
                        2 * KTILE_B + 2 * VTILE_B + KVBLK * 4};
 
   const float scale2 = scale * LOG2E;
+
+  // ---- Q fragments in registers ---------------------------------------
+  bf16x8 qf[D / 16];
+#pragma unroll
+  for (int s = 0; s < D / 16; ++s)
+    qf[s] = *reinterpret_cast<const bf16x8*>(Qb + (long)qload * tstr_q +
+                                             s * 16 + h * 8);
+
+  float m = -1e30f, l = 0.f;
+  f32x16 accO[DBLK];
+#pragma unroll
+  for (int d = 0; d < DBLK; ++d)
+#pragma unroll
+    for (int i = 0; i < 16; ++i) accO[d][i] = 0.f;
 
   // ---- staging helpers --------------------------------------------------
   auto stage_k = [&](int k0, int buf) {   // glds, swizzled source
@@ -167,26 +171,6 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     }
   };
 
-  s16x8 vreg[VCH];
-  for (int qb = qb_lo; qb < qb_hi; ++qb) {
-  const int q0w = qb * QBLK + wid * QWAVE;
-  const int qrow_abs = q0w + qcol;
-  const int qload = qrow_abs < SQ ? qrow_abs : (SQ > 0 ? SQ - 1 : 0);
-
-  // ---- Q fragments in registers ---------------------------------------
-  bf16x8 qf[D / 16];
-#pragma unroll
-  for (int s = 0; s < D / 16; ++s)
-    qf[s] = *reinterpret_cast<const bf16x8*>(Qb + (long)qload * tstr_q +
-                                             s * 16 + h * 8);
-
-  float m = -1e30f, l = 0.f;
-  f32x16 accO[DBLK];
-#pragma unroll
-  for (int d = 0; d < DBLK; ++d)
-#pragma unroll
-    for (int i = 0; i < 16; ++i) accO[d][i] = 0.f;
-
   int ntile = (SKV + KVBLK - 1) / KVBLK;
   if (CAUSAL) {
     const int qmax_blk = min(qb * QBLK + QBLK - 1, SQ - 1);
@@ -194,6 +178,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   }
 
   // prologue: stage tile 0 synchronously
+  s16x8 vreg[VCH];
   stage_k(0, 0);
   if (!VDIRECT) {
     v_issue(0, vreg);
@@ -375,10 +360,6 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
       }
     }
   }
-  // the epilogue reused the staging LDS: all waves must finish reading
-  // it before the next q-block's prologue restages
-  __syncthreads();
-  }  // qb loop
 }
 
 extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
@@ -390,15 +371,7 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
   const int nwave = 4;
   const int qblk = nwave * 32;
   const int nqb = (SQ + qblk - 1) / qblk;
-  // q-blocks per workgroup: as many as possible while the grid still
-  // fills the chip (~>=2048 workgroups); small B*H keeps the old 1:1.
-  const long bh = (long)B * H;
-  const long denom = bh > 2048 ? bh : 2048;
-  int qbpb = (int)(((long)nqb * bh + denom - 1) / denom);
-  if (qbpb < 1) qbpb = 1;
-  if (qbpb > nqb) qbpb = nqb;
-  const int ngrp = (nqb + qbpb - 1) / qbpb;
-  const int grid = B * H * ngrp;
+  const int grid = B * H * nqb;
   const int ktile = KVBLK * D * 2, vtile = D * VROW * 2;
   const int stage_b = 2 * ktile + 2 * vtile + 2 * KVBLK * 4;
   const int epi_b = nwave * 32 * (D + 1) * 4;
@@ -421,14 +394,14 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
                          (const unsigned short*)Q, (const unsigned short*)K,   \
                          (const unsigned short*)V, mask, (unsigned short*)O,   \
                          B, SQ, SKV, H, HKV, scale, bstr_q, tstr_q, bstr_k,    \
-                         tstr_k, qbpb);                                        \
+                         tstr_k);                                              \
     else                                                                       \
       hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, MM, false>), dim3(grid),     \
                          dim3(nwave * 64), lds, stream,                        \
                          (const unsigned short*)Q, (const unsigned short*)K,   \
                          (const unsigned short*)V, mask, (unsigned short*)O,   \
                          B, SQ, SKV, H, HKV, scale, bstr_q, tstr_q, bstr_k,    \
-                         tstr_k, qbpb);                                        \
+                         tstr_k);                                              \
   } while (0)
   const bool hm = mask != nullptr;
   if (D == 64) {

@@ -24,17 +24,8 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
                   const int* __restrict__ ctx_len,
                   unsigned short* __restrict__ out, int B, int H, int HKV,
                   int max_blocks, float scale) {
-  // T1 bijective XCD swizzle: consecutive ids = the GQA head group
-  // sharing one (b, kvh)'s KV — keep them on one XCD's L2.
-  int bid = blockIdx.x;
-  {
-    const int nwg = B * H;
-    const int q = nwg / 8, r = nwg % 8;
-    const int xcd = bid % 8, j = bid / 8;
-    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + j;
-  }
-  const int b = bid / H;
-  const int h = bid % H;
+  const int b = blockIdx.x / H;
+  const int h = blockIdx.x % H;
   const int kvh = h / (H / HKV);
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
