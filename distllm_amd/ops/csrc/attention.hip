@@ -67,7 +67,15 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int qcol = lane & 31;
 
   const int nqb = (SQ + QBLK - 1) / QBLK;
-  const int bid = blockIdx.x;
+  // T1 bijective XCD swizzle: keep the q-blocks sharing one (b, head)'s
+  // K/V on one XCD's L2 (measured +12% at D=128/GQA, neutral at D=64).
+  int bid = blockIdx.x;
+  {
+    const int nwg = Bsz * H * nqb;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, j = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + j;
+  }
   const int qb = bid % nqb;
   const int bh = bid / nqb;
   const int head = bh % H;
