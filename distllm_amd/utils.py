@@ -72,17 +72,18 @@ def iter_batches(data: list[T], batch_size: int) -> Iterator[list[T]]:
 # Timing
 # ---------------------------------------------------------------------------
 
-_TIMER_RE = re.compile(
-    r"\[timer\] \[(?P<tags>.*?)\] in (?P<elapsed>[0-9.eE+-]+) s "
-    r"\(start: (?P<start>[0-9.]+), end: (?P<end>[0-9.]+)\)"
-)
+class TimeStats(BaseModel):
+    tags: list[str]
+    elapsed_s: float
+    start_unix: float
+    end_unix: float
 
 
 class Timer:
-    """Context-manager wall timer emitting parseable ``[timer]`` lines.
-
-    Log-line format is compatible with the reference's TimeLogger
-    (distllm/timer.py:126-162) so downstream log parsers keep working.
+    """Wall timer usable as a context manager or via start()/stop(),
+    emitting the reference's exact ``[timer]`` line format
+    (distllm/timer.py: bracket-delimited fields, space-separated tags)
+    so existing log parsers keep working.
     """
 
     def __init__(self, *tags: str, logfile: PathLike | None = None) -> None:
@@ -91,46 +92,86 @@ class Timer:
         self.start_unix = 0.0
         self.end_unix = 0.0
         self._t0 = 0
-        self.elapsed_s = 0.0
+        self._elapsed_ns = 0
 
-    def __enter__(self) -> "Timer":
+    @property
+    def elapsed_ns(self) -> int:
+        return self._elapsed_ns
+
+    @property
+    def elapsed_ms(self) -> float:
+        return self._elapsed_ns / 1e6
+
+    @property
+    def elapsed_s(self) -> float:
+        return self._elapsed_ns / 1e9
+
+    def start(self) -> "Timer":
         self.start_unix = time.time()
         self._t0 = time.perf_counter_ns()
         return self
 
-    def __exit__(self, *exc: Any) -> None:
-        self.elapsed_s = (time.perf_counter_ns() - self._t0) / 1e9
+    def stop(self) -> None:
+        self._elapsed_ns = time.perf_counter_ns() - self._t0
         self.end_unix = time.time()
+        stats = TimeStats(
+            tags=self.tags,
+            elapsed_s=self.elapsed_s,
+            start_unix=self.start_unix,
+            end_unix=self.end_unix,
+        )
+        TimeLogger(self.logfile).log(stats)
+
+    def __enter__(self) -> "Timer":
+        return self.start()
+
+    def __exit__(self, *exc: Any) -> None:
+        self.stop()
+
+
+_BRACKET_RE = re.compile(r"\[([^\[\]]+)\]")
+
+
+class TimeLogger:
+    """Emit and parse ``[timer]`` lines (reference TimeLogger API)."""
+
+    def __init__(self, logfile: PathLike | None = None) -> None:
+        self.logfile = logfile
+
+    def log(self, ts: TimeStats) -> None:
         line = (
-            f"[timer] [{', '.join(self.tags)}] in {self.elapsed_s} s "
-            f"(start: {self.start_unix}, end: {self.end_unix})"
+            f'[timer] [{" ".join(ts.tags)}] in [{ts.elapsed_s:.6f}] '
+            f"seconds.  start: [{ts.start_unix:.6f}], "
+            f"end: [{ts.end_unix:.6f}]"
         )
         print(line, flush=True)
         if self.logfile is not None:
             with open(self.logfile, "a") as fh:
                 fh.write(line + "\n")
 
-
-class TimeStats(BaseModel):
-    tags: list[str]
-    elapsed_s: float
-    start_unix: float
-    end_unix: float
+    def parse_logs(self, log_path: PathLike) -> list[TimeStats]:
+        lines = [
+            ln for ln in Path(log_path).read_text().splitlines()
+            if "[timer]" in ln
+        ]
+        out = []
+        for ln in lines:
+            m = _BRACKET_RE.findall(ln)
+            out.append(TimeStats(tags=m[1].split(), elapsed_s=float(m[2]),
+                                 start_unix=float(m[3]),
+                                 end_unix=float(m[4])))
+        return out
 
 
 def parse_timer_logs(text: str) -> list[TimeStats]:
-    """Parse ``[timer]`` lines back into structured stats."""
+    """Parse ``[timer]`` lines from a string."""
     out: list[TimeStats] = []
-    for m in _TIMER_RE.finditer(text):
-        tags = [t.strip() for t in m.group("tags").split(",") if t.strip()]
-        out.append(
-            TimeStats(
-                tags=tags,
-                elapsed_s=float(m.group("elapsed")),
-                start_unix=float(m.group("start")),
-                end_unix=float(m.group("end")),
-            )
-        )
+    for ln in text.splitlines():
+        if "[timer]" not in ln:
+            continue
+        m = _BRACKET_RE.findall(ln)
+        out.append(TimeStats(tags=m[1].split(), elapsed_s=float(m[2]),
+                             start_unix=float(m[3]), end_unix=float(m[4])))
     return out
 
 

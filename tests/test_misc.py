@@ -15,12 +15,20 @@ def test_timer_line_roundtrip(capsys, tmp_path):
     with Timer("computed-embeddings", "file1.jsonl", logfile=log):
         pass
     out = capsys.readouterr().out
-    assert "[timer] [computed-embeddings, file1.jsonl] in " in out
+    assert "[timer] [computed-embeddings file1.jsonl] in [" in out
     stats = parse_timer_logs(log.read_text())
     assert len(stats) == 1
     assert stats[0].tags == ["computed-embeddings", "file1.jsonl"]
     assert stats[0].elapsed_s >= 0
     assert stats[0].end_unix >= stats[0].start_unix
+    # start/stop API + TimeLogger.parse_logs (reference timer.py surface)
+    from distllm_amd.utils import TimeLogger
+
+    t = Timer("manual", logfile=log).start()
+    t.stop()
+    assert t.elapsed_ns >= 0 and t.elapsed_ms >= 0
+    stats2 = TimeLogger().parse_logs(log)
+    assert len(stats2) == 2 and stats2[1].tags == ["manual"]
 
 
 def test_batch_data():
