@@ -92,14 +92,10 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int qrow_abs = q0w + qcol;
   const int qload = qrow_abs < SQ ? qrow_abs : (SQ > 0 ? SQ - 1 : 0);
 
-  // K triple-buffered (glds prefetched 2 tiles ahead); V images double
-  // (written 1 ahead) but V LOADS run 2 ahead in two register sets, so
-  // the tile-end wait is a COUNTED vmcnt and HBM latency for both
-  // streams hides under two tiles of compute.
-  const int offK[3] = {0, KTILE_B, 2 * KTILE_B};
-  const int offV[2] = {3 * KTILE_B, 3 * KTILE_B + VTILE_B};
-  const int offM[2] = {3 * KTILE_B + 2 * VTILE_B,
-                       3 * KTILE_B + 2 * VTILE_B + KVBLK * 4};
+  const int offK[2] = {0, KTILE_B};
+  const int offV[2] = {2 * KTILE_B, 2 * KTILE_B + VTILE_B};
+  const int offM[2] = {2 * KTILE_B + 2 * VTILE_B,
+                       2 * KTILE_B + 2 * VTILE_B + KVBLK * 4};
 
   const float scale2 = scale * LOG2E;
 
@@ -189,31 +185,24 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
     ntile = min(ntile, qmax_blk / KVBLK + 1);
   }
 
-  // glds instructions per wave per K tile (counted-wait arithmetic)
-  constexpr int KP = (KVBLK / (1024 / (2 * D))) / NWAVE;
-  // prologue: K tiles 0,1 staged; V tile 0 written; V tile 1 in regs
-  s16x8 vreg[2][VCH];
+  // prologue: stage tile 0 synchronously
+  s16x8 vreg[VCH];
   stage_k(0, 0);
-  if (ntile > 1) stage_k(KVBLK, 1);
   if (!VDIRECT) {
-    v_issue(0, vreg[0]);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    v_write(0, vreg[0]);
-    if (ntile > 1) v_issue(KVBLK, vreg[1]);
+    v_issue(0, vreg);
+    v_write(0, vreg);
   }
   stage_m(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
+  int cur = 0;
   for (int t = 0; t < ntile; ++t) {
     const int k0 = t * KVBLK;
-    const int cur = t & 1;             // V image / mask parity
-    const int kcur = t % 3;            // K image
     const bool have_next = t + 1 < ntile;
-    const bool have_next2 = t + 2 < ntile;
-    if (have_next2) {
-      stage_k(k0 + 2 * KVBLK, (t + 2) % 3);
-      if (!VDIRECT) v_issue(k0 + 2 * KVBLK, vreg[t & 1]);
+    if (have_next) {
+      stage_k(k0 + KVBLK, cur ^ 1);   // async glds into the other buffer
+      if (!VDIRECT) v_issue(k0 + KVBLK, vreg);  // V -> regs (T14)
     }
 
     const bool wave_active =
@@ -242,7 +231,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
           const int chunk = (s * 16 + h * 8) / 8;
           const int swz = chunk ^ (key & (NCH - 1));
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              smem + offK[kcur] + key * (2 * D) + swz * 16);
+              smem + offK[cur] + key * (2 * D) + swz * 16);
           accP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               kf, qf[s], accP, 0, 0, 0);
         }
@@ -331,34 +320,13 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
       }
     }
 
-    if (VDIRECT) {
-      // ordinary V loads share the VM counter: keep the simple drain
-      if (have_next2) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      if (have_next) stage_m(k0 + KVBLK, cur ^ 1);
-      __syncthreads();
-    } else {
-      // counted wait: drain K(t+1) + V(t+1) (issued two tiles ago and
-      // last tile), keep K(t+2) + V(t+2) in flight across the barrier.
-      if (have_next2) {
-        if constexpr (KP + VCH == 4)
-          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        else if constexpr (KP + VCH == 6)
-          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-        else
-          asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      if (have_next) {
-        v_write(cur ^ 1, vreg[(t + 1) & 1]);
-        stage_m(k0 + KVBLK, cur ^ 1);
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
-      asm volatile("" ::: "memory");
+    if (have_next) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // V regs + glds
+      if (!VDIRECT) v_write(cur ^ 1, vreg);
+      stage_m(k0 + KVBLK, cur ^ 1);
     }
+    __syncthreads();  // ONE barrier per tile
+    cur ^= 1;
   }
 
   // ---- epilogue: O^T regs -> padded LDS -> coalesced bf16 stores ------
@@ -413,7 +381,7 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
   const int nqb = (SQ + qblk - 1) / qblk;
   const int grid = B * H * nqb;
   const int ktile = KVBLK * D * 2, vtile = D * VROW * 2;
-  const int stage_b = 3 * ktile + 2 * vtile + 2 * KVBLK * 4;
+  const int stage_b = 2 * ktile + 2 * vtile + 2 * KVBLK * 4;
   const int epi_b = nwave * 32 * (D + 1) * 4;
   const int lds = stage_b > epi_b ? stage_b : epi_b;
 
