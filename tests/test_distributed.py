@@ -88,3 +88,37 @@ def test_tensor_parallel_two_ranks(tmp_path):
     r = _torchrun(["tests/tp_worker.py", str(tmp_path)], timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "TP_OK" in r.stdout, r.stdout
+
+
+def test_distributed_generation_two_ranks(tmp_path):
+    idir = tmp_path / "in"
+    idir.mkdir()
+    for i in range(2):
+        with open(idir / f"f{i}.jsonl", "w") as fh:
+            for j in range(2):
+                fh.write(json.dumps({"text": f"Prompt {i} {j} here",
+                                     "path": f"p{i}"}) + "\n")
+    odir = tmp_path / "out"
+    cfg = {
+        "input_dir": str(idir),
+        "output_dir": str(odir),
+        "glob_patterns": ["*.jsonl"],
+        "prompt_config": {"name": "identity"},
+        "reader_config": {"name": "jsonl"},
+        "writer_config": {"name": "huggingface"},
+        "generator_config": {"name": "native", "model": "llama-tiny",
+                             "temperature": 0.0, "min_p": 0.0,
+                             "max_tokens": 3, "max_model_len": 128,
+                             "max_batch_size": 2},
+    }
+    cfg_path = tmp_path / "g.yaml"
+    cfg_path.write_text(yaml.safe_dump(cfg))
+    r = _torchrun(["-m", "distllm_amd.distributed_generation",
+                   "--config", str(cfg_path)])
+    assert r.returncode == 0, r.stdout + r.stderr
+    shards = list((odir / "generations").iterdir())
+    assert len(shards) == 2
+    from datasets import load_from_disk
+
+    total = sum(len(load_from_disk(str(s))) for s in shards)
+    assert total == 4

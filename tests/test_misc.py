@@ -125,3 +125,49 @@ def test_model_weight_loader_roundtrip(tmp_path):
     load_weights(m2, tmp_path)
     assert torch.equal(m1.word_emb, m2.word_emb)
     assert torch.equal(m1.layers[0].wqkv, m2.layers[0].wqkv)
+
+
+def test_esm2_weight_mapping(tmp_path):
+    """HF EsmModel-named checkpoint loads into our Esm2Encoder."""
+    from safetensors.torch import save_file
+
+    from distllm_amd.models import build_model
+    from distllm_amd.models.loader import load_weights
+
+    m1 = build_model("esm2-8m", dtype=torch.float32)
+    sd = {}
+    sd["esm.embeddings.word_embeddings.weight"] = m1.word_emb
+    sd["esm.encoder.emb_layer_norm_after.weight"] = m1.final_ln_w
+    sd["esm.encoder.emb_layer_norm_after.bias"] = m1.final_ln_b
+    H = m1.dims.hidden_size
+    for i, lyr in enumerate(m1.layers):
+        p = f"esm.encoder.layer.{i}."
+        q, k, v = lyr.wqkv[:H], lyr.wqkv[H:2 * H], lyr.wqkv[2 * H:]
+        qb, kb, vb = lyr.bqkv[:H], lyr.bqkv[H:2 * H], lyr.bqkv[2 * H:]
+        sd[p + "attention.self.query.weight"] = q
+        sd[p + "attention.self.key.weight"] = k
+        sd[p + "attention.self.value.weight"] = v
+        sd[p + "attention.self.query.bias"] = qb
+        sd[p + "attention.self.key.bias"] = kb
+        sd[p + "attention.self.value.bias"] = vb
+        sd[p + "attention.output.dense.weight"] = lyr.wo
+        sd[p + "attention.output.dense.bias"] = lyr.bo
+        sd[p + "attention.LayerNorm.weight"] = lyr.ln1_w
+        sd[p + "attention.LayerNorm.bias"] = lyr.ln1_b
+        sd[p + "LayerNorm.weight"] = lyr.ln2_w
+        sd[p + "LayerNorm.bias"] = lyr.ln2_b
+        sd[p + "intermediate.dense.weight"] = lyr.w1
+        sd[p + "intermediate.dense.bias"] = lyr.b1
+        sd[p + "output.dense.weight"] = lyr.w2
+        sd[p + "output.dense.bias"] = lyr.b2
+    save_file({k: v.detach().clone().contiguous() for k, v in sd.items()},
+              str(tmp_path / "model.safetensors"))
+    m2 = build_model("esm2-8m", dtype=torch.float32)
+    load_weights(m2, tmp_path)
+    assert torch.equal(m1.word_emb, m2.word_emb)
+    assert torch.equal(m1.layers[0].wqkv, m2.layers[0].wqkv)
+    # forwards agree
+    ids = torch.randint(4, 25, (1, 12))
+    mask = torch.ones_like(ids, dtype=torch.int32)
+    with torch.no_grad():
+        assert torch.allclose(m1(ids, mask), m2(ids, mask))
