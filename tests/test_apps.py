@@ -172,3 +172,38 @@ def test_cli_chunk_fasta(tmp_path):
     ])
     assert r.exit_code == 0, r.output
     assert len(list(out.glob("*.fasta"))) == 2
+
+
+def test_prompt_postprocess_behaviors():
+    from distllm_amd.generate import get_prompt_template
+
+    qc = get_prompt_template({"name": "question_chunk"})
+    out = qc.postprocess([
+        "Here is context. What drives protein folding? It is known.",
+        "No questions at all here.",
+    ])
+    assert out[0].strip().endswith("?")
+    assert "folding" in out[0]
+    assert out[1] == ""
+
+    qa = get_prompt_template({"name": "question_answer"})
+    assert qa.postprocess(["2. Mitochondria"]) == ["mitochondria"]
+    assert qa.postprocess(["Option 3: DNA"]) == ["dna"]
+    prompts = qa.preprocess(["Q?"], [["ctx a", "ctx b"]], [[0.9, 0.5]])
+    assert "score: 0.9" in prompts[0] and "Q?" in prompts[0]
+
+    ident = get_prompt_template({"name": "identity"})
+    assert ident.preprocess("solo") == ["solo"]
+
+    ks = get_prompt_template({"name": "keyword_selection",
+                              "keywords": ["radiation", "dosimetry"]})
+    p = ks.preprocess(["A paragraph."])[0]
+    assert "radiation" in p and "A paragraph." in p
+
+    amp = get_prompt_template({"name": "amp_question"})
+    p2 = amp.preprocess(['{"name": "pep1", "function": "antimicrobial"}'])
+    assert "pep1" in p2[0]
+    import json as _json
+
+    post = amp.postprocess(["Q: which peptide?"])
+    assert _json.loads(post[0])["question"] == "Q: which peptide?"
