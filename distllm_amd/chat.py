@@ -97,14 +97,21 @@ class ChatSession:
         self.last_retrieval: list[tuple[float, str]] = []
         self.batcher = None
         if batching:
-            from .engine.batcher import DynamicBatcher
+            if hasattr(generator, "generate_one"):
+                # native engine: request-level continuous batching
+                generator.get_loop()
+            else:
+                from .engine.batcher import DynamicBatcher
 
-            self.batcher = DynamicBatcher(generator)
+                self.batcher = DynamicBatcher(generator)
 
     def _generate(self, prompt: str) -> str:
+        gen = self.rag.generator
+        if hasattr(gen, "generate_one") and getattr(gen, "_loop", None):
+            return gen.generate_one(prompt)
         if self.batcher is not None:
             return self.batcher.generate_one(prompt)
-        return self.rag.generator.generate([prompt])[0]
+        return gen.generate([prompt])[0]
 
     def answer(self, history: list[tuple[str, str]], user_text: str
                ) -> tuple[str, list[tuple[float, str]]]:

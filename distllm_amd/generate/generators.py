@@ -82,6 +82,7 @@ class NativeGenerator:
         self.engine = LLMEngine(self.model,
                                 max_batch_size=config.max_batch_size,
                                 max_model_len=config.max_model_len)
+        self._loop = None
         eos = getattr(self.tokenizer, "eos_token_id", None)
         self.params = SamplingParams(
             temperature=config.temperature,
@@ -91,23 +92,42 @@ class NativeGenerator:
             stop_token_ids=(eos,) if eos is not None else (),
         )
 
-    def generate(self, prompts: list[str]) -> list[str]:
-        if isinstance(prompts, str):
-            prompts = [prompts]
+    def get_loop(self):
+        """Online engine loop (request-level continuous batching);
+        once created it owns the engine and all generate() calls route
+        through it (thread-safe)."""
+        if self._loop is None:
+            from ..engine.loop import EngineLoop
+
+            self._loop = EngineLoop(self.engine, self.params)
+        return self._loop
+
+    def generate_one(self, prompt: str) -> str:
+        loop = self.get_loop()
+        ids = self._tokenize([prompt])[0]
+        out = loop.generate([ids])[0]
+        return self.tokenizer.decode(out, skip_special_tokens=True)
+
+    def _tokenize(self, prompts: list[str]) -> list[list[int]]:
         enc = self.tokenizer(prompts, padding=False, truncation=True,
                              return_tensors=None)
         ids = enc["input_ids"]
-        # tokenizer may return tensors or lists depending on provider
         import torch
 
         if isinstance(ids, torch.Tensor):
             mask = enc["attention_mask"]
-            prompt_ids = [
-                ids[i][mask[i].bool()].tolist() for i in range(len(prompts))
-            ]
+            return [ids[i][mask[i].bool()].tolist()
+                    for i in range(len(prompts))]
+        return [list(x) for x in ids]
+
+    def generate(self, prompts: list[str]) -> list[str]:
+        if isinstance(prompts, str):
+            prompts = [prompts]
+        prompt_ids = self._tokenize(prompts)
+        if self._loop is not None:  # loop owns the engine once started
+            out_ids = self._loop.generate(prompt_ids)
         else:
-            prompt_ids = [list(x) for x in ids]
-        out_ids = self.engine.generate(prompt_ids, self.params)
+            out_ids = self.engine.generate(prompt_ids, self.params)
         return [
             self.tokenizer.decode(o, skip_special_tokens=True)
             for o in out_ids

@@ -292,10 +292,12 @@ def rope_apply(x, cs, sn, pos):
 rope_tables = torch_ref.rope_tables
 
 
-def topk(scores, k, chunk=1 << 18):
+def topk(scores, k, chunk=1 << 15):
     """Row-wise top-k over a [Q, N] f32 score matrix.
 
-    GPU: chunked radix-select kernel + candidate-merge round.
+    GPU: chunked radix-select kernel + candidate-merge round. The
+    32k default chunk keeps the grid >= 1024 workgroups even for a few
+    queries (the chip needs >> 256 workgroups to fill).
     Returns (values [Q,k] f32 desc-sorted, indices [Q,k] int32).
     """
     if scores.is_cuda:
