@@ -82,7 +82,8 @@ class LLMEngine:
                  max_model_len: int = 4096,
                  kv_memory_fraction: float = 0.6,
                  num_blocks: int | None = None,
-                 use_graphs: bool | None = None):
+                 use_graphs: bool | None = None,
+                 kv_cache_dtype: str = "bf16"):
         self.model = model
         d = model.dims
         self.device = model.device
@@ -91,7 +92,13 @@ class LLMEngine:
         dev = self.device
         hkv = getattr(model, 'local_kv_heads', d.kv_heads)
         hd, L = d.head_dim, d.num_layers
-        bytes_per_block_layer = hkv * self.BLK * hd * 2 * 2  # k+v bf16
+        # optional OCP fp8-e4m3 KV cache (CDNA4-native): halves KV
+        # bandwidth/footprint; off by default (bf16 = reference dtype)
+        if kv_cache_dtype not in ("bf16", "fp8_e4m3"):
+            raise ValueError(f"kv_cache_dtype {kv_cache_dtype!r}")
+        self.kv_cache_dtype = kv_cache_dtype
+        kv_el = 1 if kv_cache_dtype == "fp8_e4m3" else 2
+        bytes_per_block_layer = hkv * self.BLK * hd * kv_el * 2  # k+v
         if num_blocks is None:
             if str(dev).startswith("cuda"):
                 free, _total = torch.cuda.mem_get_info(dev)
@@ -104,8 +111,12 @@ class LLMEngine:
             num_blocks = min(num_blocks, need)
         num_blocks += 1  # block 0 is reserved (graph-padding dummy)
         self.num_blocks = num_blocks
+        cache_dt = (torch.uint8 if kv_cache_dtype == "fp8_e4m3"
+                    else model.dtype)
+        if kv_cache_dtype == "fp8_e4m3" and not str(dev).startswith("cuda"):
+            raise ValueError("fp8_e4m3 KV cache requires a GPU")
         self.kcaches = [
-            torch.zeros(num_blocks, hkv, self.BLK, hd, dtype=model.dtype,
+            torch.zeros(num_blocks, hkv, self.BLK, hd, dtype=cache_dt,
                         device=dev)
             for _ in range(L)
         ]

@@ -38,6 +38,7 @@ class NativeGeneratorConfig(BaseConfig):
     tensor_parallel_size: int = 1
     trust_remote_code: bool = True
     quantization: bool = False
+    kv_cache_dtype: str = "bf16"  # or "fp8_e4m3" (CDNA4 OCP fp8)
 
 
 class NativeGenerator:
@@ -81,7 +82,8 @@ class NativeGenerator:
         )
         self.engine = LLMEngine(self.model,
                                 max_batch_size=config.max_batch_size,
-                                max_model_len=config.max_model_len)
+                                max_model_len=config.max_model_len,
+                                kv_cache_dtype=config.kv_cache_dtype)
         self._loop = None
         eos = getattr(self.tokenizer, "eos_token_id", None)
         self.params = SamplingParams(

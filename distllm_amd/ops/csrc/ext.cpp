@@ -50,9 +50,9 @@ void launch_sample(const float*, const float*, int*, int, int, float, float,
                    hipStream_t);
 void launch_paged_attn(const void*, const void*, const void*, const int*,
                        const int*, void*, int, int, int, int, int, float,
-                       hipStream_t);
+                       bool, hipStream_t);
 void launch_cache_put(const void*, const void*, void*, void*, const int*,
-                      const int*, int, int, int, int, hipStream_t);
+                      const int*, int, int, int, int, bool, hipStream_t);
 }
 
 // ---- norms ----------------------------------------------------------------
@@ -297,11 +297,12 @@ torch::Tensor paged_attn(torch::Tensor q, torch::Tensor kcache,
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int HKV = kcache.size(1);
   const int max_blocks = block_table.size(1);
+  const bool fp8 = kcache.scalar_type() == at::kByte;
   auto out = torch::empty_like(q);
   launch_paged_attn(q.data_ptr(), kcache.data_ptr(), vcache.data_ptr(),
                     block_table.data_ptr<int>(), ctx_len.data_ptr<int>(),
                     out.data_ptr(), B, H, HKV, D, max_blocks, (float)scale,
-                    cur_stream());
+                    fp8, cur_stream());
   return out;
 }
 
@@ -310,10 +311,11 @@ void cache_put(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kcache,
                torch::Tensor slot) {
   CHECK_DEV(knew); CHECK_CONT(knew); CHECK_BF16(knew);
   const int B = knew.size(0), HKV = knew.size(1), D = knew.size(2);
+  const bool fp8 = kcache.scalar_type() == at::kByte;
   launch_cache_put(knew.data_ptr(), vnew.data_ptr(), kcache.data_ptr(),
                    vcache.data_ptr(), block_table.data_ptr<int>(),
                    slot.data_ptr<int>(), B, HKV, D, block_table.size(1),
-                   cur_stream());
+                   fp8, cur_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

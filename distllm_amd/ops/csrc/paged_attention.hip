@@ -12,14 +12,24 @@
 // 8-deep unroll so the per-key V reads pipeline instead of serializing
 // on L2 latency. Cross-wave combine via LDS.
 #include "common.h"
+#include <hip/hip_fp8.h>
 
 #define PBLK 16  // tokens per KV block
 
-template <int D>
+__device__ __forceinline__ float fp8_to_f32(unsigned char u) {
+  __hip_fp8_e4m3 v;
+  v.__x = u;
+  return float(v);
+}
+
+// CT: cache element type (unsigned short = bf16, unsigned char = OCP
+// fp8 e4m3 — CDNA4-native; halves decode KV bandwidth, doubles
+// capacity; values are stored unscaled, |k|,|v| << 448 in practice).
+template <int D, typename CT>
 __global__ void __launch_bounds__(256)
 paged_attn_kernel(const unsigned short* __restrict__ q,
-                  const unsigned short* __restrict__ kcache,
-                  const unsigned short* __restrict__ vcache,
+                  const CT* __restrict__ kcache,
+                  const CT* __restrict__ vcache,
                   const int* __restrict__ block_table,
                   const int* __restrict__ ctx_len,
                   unsigned short* __restrict__ out, int B, int H, int HKV,
@@ -55,15 +65,23 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
     const int key = t0 + lane;
     float s = -INFINITY;
     if (key < L) {
-      const unsigned short* krow = kcache +
+      const CT* krow = kcache +
           (long)bt_lds[key / PBLK] * blk_stride + head_off +
           (key % PBLK) * D;
       float dot = 0.f;
 #pragma unroll
       for (int c = 0; c < D / 8; ++c) {
-        s16x8 kv8 = *reinterpret_cast<const s16x8*>(krow + c * 8);
         float kf[8];
-        bf8_to_f32(kv8, kf);
+        if constexpr (sizeof(CT) == 2) {
+          s16x8 kv8 = *reinterpret_cast<const s16x8*>(krow + c * 8);
+          bf8_to_f32(kv8, kf);
+        } else {
+          i32x2 kv8 = *reinterpret_cast<const i32x2*>(krow + c * 8);
+          const unsigned char* b =
+              reinterpret_cast<const unsigned char*>(&kv8);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kf[i] = fp8_to_f32(b[i]);
+        }
 #pragma unroll
         for (int i = 0; i < 8; ++i) dot += kf[i] * q_lds[c * 8 + i];
       }
@@ -89,11 +107,14 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
       for (int j = 0; j < 8; ++j) {
         const int key2 = t0 + kk + j;
         pv[j] = p_lds[wid * 64 + kk + j];
-        const unsigned short* vrow = vcache +
+        const CT* vrow = vcache +
             (long)bt_lds[key2 / PBLK] * blk_stride + head_off +
             (key2 % PBLK) * D + lane * DPL;
 #pragma unroll
-        for (int i = 0; i < DPL; ++i) vv[j][i] = bf2f(vrow[i]);
+        for (int i = 0; i < DPL; ++i)
+          vv[j][i] = (sizeof(CT) == 2)
+              ? bf2f((unsigned short)vrow[i])
+              : fp8_to_f32((unsigned char)vrow[i]);
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -103,11 +124,14 @@ paged_attn_kernel(const unsigned short* __restrict__ q,
     for (; kk < kmax; ++kk) {
       const int key2 = t0 + kk;
       const float pv = p_lds[wid * 64 + kk];
-      const unsigned short* vrow = vcache +
+      const CT* vrow = vcache +
           (long)bt_lds[key2 / PBLK] * blk_stride + head_off +
           (key2 % PBLK) * D + lane * DPL;
 #pragma unroll
-      for (int i = 0; i < DPL; ++i) acc[i] += pv * bf2f(vrow[i]);
+      for (int i = 0; i < DPL; ++i)
+        acc[i] += pv * ((sizeof(CT) == 2)
+                        ? bf2f((unsigned short)vrow[i])
+                        : fp8_to_f32((unsigned char)vrow[i]));
     }
   }
 
@@ -143,29 +167,30 @@ extern "C" void launch_paged_attn(const void* q, const void* kcache,
                                   const void* vcache, const int* block_table,
                                   const int* ctx_len, void* out, int B, int H,
                                   int HKV, int D, int max_blocks, float scale,
-                                  hipStream_t stream) {
+                                  bool fp8, hipStream_t stream) {
   const int lds = (4 * 64 + D + 8 + 4 * D) * 4 + max_blocks * 4;
   dim3 grid(B * H);
-  if (D == 64)
-    hipLaunchKernelGGL((paged_attn_kernel<64>), grid, dim3(256), lds, stream,
-                       (const unsigned short*)q, (const unsigned short*)kcache,
-                       (const unsigned short*)vcache, block_table, ctx_len,
-                       (unsigned short*)out, B, H, HKV, max_blocks, scale);
-  else if (D == 128)
-    hipLaunchKernelGGL((paged_attn_kernel<128>), grid, dim3(256), lds, stream,
-                       (const unsigned short*)q, (const unsigned short*)kcache,
-                       (const unsigned short*)vcache, block_table, ctx_len,
-                       (unsigned short*)out, B, H, HKV, max_blocks, scale);
+#define PA(DD, CT)                                                             \
+  hipLaunchKernelGGL((paged_attn_kernel<DD, CT>), grid, dim3(256), lds,        \
+                     stream, (const unsigned short*)q, (const CT*)kcache,      \
+                     (const CT*)vcache, block_table, ctx_len,                  \
+                     (unsigned short*)out, B, H, HKV, max_blocks, scale)
+  if (D == 64) {
+    if (fp8) PA(64, unsigned char); else PA(64, unsigned short);
+  } else if (D == 128) {
+    if (fp8) PA(128, unsigned char); else PA(128, unsigned short);
+  }
+#undef PA
 }
 
 // Write one new token's K/V into the paged cache (decode step).
 // knew/vnew: [B, HKV, D] bf16; slot[b] = ctx position of the new token.
-template <int D>
+template <int D, typename CT>
 __global__ void __launch_bounds__(256)
 cache_put_kernel(const unsigned short* __restrict__ knew,
                  const unsigned short* __restrict__ vnew,
-                 unsigned short* __restrict__ kcache,
-                 unsigned short* __restrict__ vcache,
+                 CT* __restrict__ kcache,
+                 CT* __restrict__ vcache,
                  const int* __restrict__ block_table,
                  const int* __restrict__ slot, int B, int HKV,
                  int max_blocks) {
@@ -181,8 +206,13 @@ cache_put_kernel(const unsigned short* __restrict__ knew,
     const int blk = block_table[(long)b * max_blocks + pos / PBLK];
     const long off =
         (long)blk * blk_stride + ((long)kvh * PBLK + pos % PBLK) * D + d;
-    kcache[off] = knew[i];
-    vcache[off] = vnew[i];
+    if constexpr (sizeof(CT) == 2) {
+      kcache[off] = knew[i];
+      vcache[off] = vnew[i];
+    } else {
+      kcache[off] = __hip_fp8_e4m3(bf2f(knew[i])).__x;
+      vcache[off] = __hip_fp8_e4m3(bf2f(vnew[i])).__x;
+    }
   }
 }
 
@@ -190,16 +220,18 @@ extern "C" void launch_cache_put(const void* knew, const void* vnew,
                                  void* kcache, void* vcache,
                                  const int* block_table, const int* slot,
                                  int B, int HKV, int D, int max_blocks,
-                                 hipStream_t stream) {
+                                 bool fp8, hipStream_t stream) {
   const long total = (long)B * HKV * D;
   int blocks = (int)min((total + 255) / 256, (long)2048);
-#define CP(DD)                                                                 \
-  hipLaunchKernelGGL((cache_put_kernel<DD>), dim3(blocks), dim3(256), 0,       \
+#define CP(DD, CT)                                                             \
+  hipLaunchKernelGGL((cache_put_kernel<DD, CT>), dim3(blocks), dim3(256), 0,   \
                      stream, (const unsigned short*)knew,                      \
-                     (const unsigned short*)vnew, (unsigned short*)kcache,     \
-                     (unsigned short*)vcache, block_table, slot, B, HKV,       \
-                     max_blocks)
-  if (D == 64) CP(64);
-  else CP(128);
+                     (const unsigned short*)vnew, (CT*)kcache, (CT*)vcache,    \
+                     block_table, slot, B, HKV, max_blocks)
+  if (D == 64) {
+    if (fp8) CP(64, unsigned char); else CP(64, unsigned short);
+  } else {
+    if (fp8) CP(128, unsigned char); else CP(128, unsigned short);
+  }
 #undef CP
 }

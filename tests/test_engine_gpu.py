@@ -56,3 +56,19 @@ def test_engine_sampling_runs_gpu(model):
     assert len(outs) == 8
     for o in outs:
         assert all(0 <= t < model.dims.vocab_size for t in o)
+
+
+def test_fp8_kv_cache_decode(model):
+    """fp8-e4m3 KV cache: greedy decode stays close to bf16 KV."""
+    engine16 = LLMEngine(model, max_batch_size=2, max_model_len=128,
+                         num_blocks=64)
+    engine8 = LLMEngine(model, max_batch_size=2, max_model_len=128,
+                        num_blocks=64, kv_cache_dtype="fp8_e4m3")
+    assert engine8.kcaches[0].dtype == torch.uint8
+    params = SamplingParams(temperature=0.0, min_p=0.0, max_tokens=8)
+    prompts = [[5, 9, 100, 42, 8, 1], list(range(20, 44))]
+    o16 = engine16.generate(prompts, params)
+    o8 = engine8.generate(prompts, params)
+    agree = sum(a == b for x, y in zip(o16, o8) for a, b in zip(x, y))
+    total = sum(len(x) for x in o16)
+    assert agree / total > 0.6, f"fp8 KV diverged: {o8} vs {o16}"
