@@ -216,10 +216,15 @@ class LLMEngine:
         bsel, tsel = valid.nonzero(as_tuple=True)
         flat_blk = blk_idx[bsel, tsel]
         flat_slot = slot[bsel, tsel]
+        fp8 = self.kv_cache_dtype == "fp8_e4m3"
         for li, (k, v) in enumerate(kv):
             # k/v: [B, maxlen, HKV, D]
-            self.kcaches[li][flat_blk, :, flat_slot] = k[bsel, tsel]
-            self.vcaches[li][flat_blk, :, flat_slot] = v[bsel, tsel]
+            ksel, vsel = k[bsel, tsel], v[bsel, tsel]
+            if fp8:  # OCP e4m3fn bit pattern == the kernel's fp8 type
+                ksel = ksel.to(torch.float8_e4m3fn).view(torch.uint8)
+                vsel = vsel.to(torch.float8_e4m3fn).view(torch.uint8)
+            self.kcaches[li][flat_blk, :, flat_slot] = ksel
+            self.vcaches[li][flat_blk, :, flat_slot] = vsel
         # last-token logits -> first sampled token
         last = (lens - 1).long()
         hlast = hidden[torch.arange(B, device=dev), last]
