@@ -192,7 +192,10 @@ def linear(x, weight, bias=None, act=ACT_NONE):
         mode = os.environ.get("DISTLLM_GEMM", "auto")
         if (mode != "blaslt" and act == ACT_NONE and M <= 256
                 and N % 64 == 0 and K % 64 == 0 and x2.is_contiguous()):
-            key = ("dec", M, N, K, bias is not None)
+            # bucket by padded M so varying decode/prefill
+            # batch sizes reuse one tuning decision
+            key = ("dec", 128 if M <= 128 else 256, N, K,
+                   bias is not None)
             choice = _GEMM_CHOICE.get(key)
             sk = _decode_splitk(M, N, K)
             if choice is None:
