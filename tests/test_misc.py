@@ -171,3 +171,42 @@ def test_esm2_weight_mapping(tmp_path):
     mask = torch.ones_like(ids, dtype=torch.int32)
     with torch.no_grad():
         assert torch.allclose(m1(ids, mask), m2(ids, mask))
+
+
+def test_example_configs_parse():
+    """The shipped example YAMLs must parse into their config models."""
+    import yaml
+
+    from distllm_amd.chat import ChatAppConfig
+    from distllm_amd.distributed_embedding import EmbeddingConfig
+    from distllm_amd.distributed_generation import GenerationConfig
+    from distllm_amd.mcqa.harness import MCQAConfig
+    from distllm_amd.rag.evaluate import EvalSuiteConfig
+
+    EmbeddingConfig(**yaml.safe_load(open("examples/embed.yaml")))
+    GenerationConfig(**yaml.safe_load(open("examples/generate.yaml")))
+    ChatAppConfig(**yaml.safe_load(open("examples/chat.yaml")))
+    MCQAConfig(**yaml.safe_load(open("examples/mcqa.yaml")))
+    EvalSuiteConfig(**yaml.safe_load(open("examples/evaluate.yaml")))
+
+
+def test_cli_generate(tmp_path):
+    from typer.testing import CliRunner
+
+    from distllm_amd.cli import app
+
+    runner = CliRunner()
+    idir = tmp_path / "in"
+    idir.mkdir()
+    with open(idir / "a.jsonl", "w") as fh:
+        fh.write(json.dumps({"text": "A prompt here", "path": "p0"}) + "\n")
+    odir = tmp_path / "gen"
+    r = runner.invoke(app, [
+        "generate", "--input-dir", str(idir), "--output-dir", str(odir),
+        "--prompt-name", "identity", "--reader-name", "jsonl",
+        "--llm-name", "llama-tiny", "--max-tokens", "3",
+        "--temperature", "0.0", "--min-p", "0.0",
+    ])
+    assert r.exit_code == 0, r.output
+    shards = [p for p in odir.iterdir() if p.is_dir()]
+    assert len(shards) == 1
