@@ -99,17 +99,12 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
 
   const float scale2 = scale * LOG2E;
 
-  // ---- Q fragments -----------------------------------------------------
-  // LDS-V path (D=64): Q cached in registers (cheap at 4 frags).
-  // VDIRECT path (D=128): re-load per QK step from L1 (the row stays
-  // hot) — frees ~50 VGPRs, which lifts the kernel to 2 waves/SIMD.
-  const unsigned short* qrow = Qb + (long)qload * tstr_q + h * 8;
-  bf16x8 qf[VDIRECT ? 1 : D / 16];
-  if (!VDIRECT) {
+  // ---- Q fragments in registers ---------------------------------------
+  bf16x8 qf[D / 16];
 #pragma unroll
-    for (int s = 0; s < D / 16; ++s)
-      qf[s] = *reinterpret_cast<const bf16x8*>(qrow + s * 16);
-  }
+  for (int s = 0; s < D / 16; ++s)
+    qf[s] = *reinterpret_cast<const bf16x8*>(Qb + (long)qload * tstr_q +
+                                             s * 16 + h * 8);
 
   float m = -1e30f, l = 0.f;
   f32x16 accO[DBLK];
@@ -237,11 +232,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
           const int swz = chunk ^ (key & (NCH - 1));
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
               smem + offK[cur] + key * (2 * D) + swz * 16);
-          bf16x8 qv = VDIRECT
-              ? *reinterpret_cast<const bf16x8*>(qrow + s * 16)
-              : qf[VDIRECT ? 0 : s];
           accP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              kf, qv, accP, 0, 0, 0);
+              kf, qf[s], accP, 0, 0, 0);
         }
         __builtin_amdgcn_s_setprio(0);
 #pragma unroll
