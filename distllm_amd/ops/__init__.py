@@ -146,13 +146,25 @@ def _blaslt_path(x2, weight, bias, act):
     return out
 
 
+def _blt_fused_path(x2, weight, bias, act):
+    """ONE hipBLASLt call with the bias/GELU/swish applied in the GEMM
+    epilogue — no separate elementwise pass over the [M, N] output."""
+    epi = (1 if act in (ACT_GELU_TANH, ACT_GELU_ERF)
+           else 2 if act == ACT_SILU else 0)
+    return _hip_ops.blt_linear(x2, weight, bias, epi)
+
+
 def _autotune_gemm(x2, weight, bias, act):
-    """Measure hipBLASLt(+epilogue kernel) vs the fused gemm8 MFMA
-    kernel once per (M-bucket, N, K, act) and cache the winner."""
+    """Measure hipBLASLt(+epilogue kernel) vs hipBLASLt-fused-epilogue
+    vs the fused gemm8 MFMA kernel once per (M-bucket, N, K, act) and
+    cache the winner."""
     import time as _t
 
     def timeit(fn):
-        fn()
+        try:
+            fn()
+        except RuntimeError:
+            return float("inf")
         torch.cuda.synchronize()
         t0 = _t.perf_counter()
         for _ in range(5):
@@ -160,9 +172,12 @@ def _autotune_gemm(x2, weight, bias, act):
         torch.cuda.synchronize()
         return _t.perf_counter() - t0
 
-    t_blas = timeit(lambda: _blaslt_path(x2, weight, bias, act))
-    t_hip = timeit(lambda: _hip_ops.gemm8_bf16(x2, weight, bias, act))
-    return "gemm8" if t_hip < t_blas else "blaslt"
+    times = {
+        "blaslt": timeit(lambda: _blaslt_path(x2, weight, bias, act)),
+        "gemm8": timeit(lambda: _hip_ops.gemm8_bf16(x2, weight, bias, act)),
+        "blt_fused": timeit(lambda: _blt_fused_path(x2, weight, bias, act)),
+    }
+    return min(times, key=times.get)
 
 
 def _decode_splitk(M, N, K):
@@ -232,9 +247,12 @@ def linear(x, weight, bias=None, act=ACT_NONE):
                 if choice is None:
                     choice = _autotune_gemm(x2, weight, bias, act)
                     _GEMM_CHOICE[key] = choice
-                out = (_hip_ops.gemm8_bf16(x2, weight, bias, act)
-                       if choice == "gemm8"
-                       else _blaslt_path(x2, weight, bias, act))
+                if choice == "gemm8":
+                    out = _hip_ops.gemm8_bf16(x2, weight, bias, act)
+                elif choice == "blt_fused":
+                    out = _blt_fused_path(x2, weight, bias, act)
+                else:
+                    out = _blaslt_path(x2, weight, bias, act)
         else:
             out = _blaslt_path(x2, weight, bias, act)
     else:

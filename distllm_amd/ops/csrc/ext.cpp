@@ -318,7 +318,13 @@ void cache_put(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kcache,
                    fp8, cur_stream());
 }
 
+// blt_gemm.cpp: hipBLASLt GEMM with fused bias/GELU/swish epilogue.
+torch::Tensor blt_linear(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias, int64_t epi);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("blt_linear", &blt_linear,
+        "hipBLASLt linear with fused bias/GELU/swish epilogue");
   m.def("norm_fwd", &norm_fwd, "fused residual+LayerNorm/RMSNorm");
   m.def("bias_gelu", &bias_gelu);
   m.def("swiglu", &swiglu);

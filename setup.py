@@ -16,7 +16,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 CSRC = Path(__file__).parent / "distllm_amd" / "ops" / "csrc"
 # *_hip.hip files are torch-hipify build artifacts — exclude them.
-SOURCES = [str(CSRC / "ext.cpp")] + sorted(
+SOURCES = [str(CSRC / "ext.cpp"), str(CSRC / "blt_gemm.cpp")] + sorted(
     str(p) for p in CSRC.glob("*.hip") if not p.name.endswith("_hip.hip")
 )
 
@@ -30,6 +30,7 @@ setup(
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950", "-std=c++17"],
             },
+            libraries=["hipblaslt"],
         )
     ],
     cmdclass={"build_ext": BuildExtension},

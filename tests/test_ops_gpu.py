@@ -256,3 +256,24 @@ def test_decode_gemm(device, M, N, K, sk):
     out = ops._hip_ops.decode_gemm(x, w, b, sk)
     ref = torch_ref.gemm(x, w, b)
     _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("act,bias", [
+    (ops.ACT_NONE, True), (ops.ACT_NONE, False),
+    (ops.ACT_GELU_ERF, True), (ops.ACT_GELU_TANH, True),
+    (ops.ACT_SILU, False),
+])
+def test_blt_fused_epilogue(device, act, bias):
+    """hipBLASLt fused bias/GELU/swish epilogue vs fp32 reference.
+    (hipBLASLt's GELU may be the tanh approximation; its max deviation
+    from erf-GELU is ~3e-3, inside bf16 tolerance.)"""
+    torch.manual_seed(11)
+    M, N, K = 512, 768, 512
+    x = torch.randn(M, K, device=device, dtype=torch.bfloat16)
+    w = torch.randn(N, K, device=device, dtype=torch.bfloat16) * 0.05
+    b = (torch.randn(N, device=device, dtype=torch.bfloat16)
+         if bias else None)
+    from distllm_amd.ops import _blt_fused_path
+    out = _blt_fused_path(x, w, b, act)
+    ref = torch_ref.gemm(x, w, b, act=act)
+    _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
