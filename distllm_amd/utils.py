@@ -109,11 +109,27 @@ class Timer:
     def start(self) -> "Timer":
         self.start_unix = time.time()
         self._t0 = time.perf_counter_ns()
+        # rocprof-visible named range (torch.cuda.nvtx emits rocTX
+        # markers on ROCm): timed stages show up in rocprofv3
+        # marker-domain traces under the same tag string.
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                torch.cuda.nvtx.range_push(" ".join(self.tags) or "timer")
+                self._nvtx = True
+        except Exception:
+            pass
         return self
 
     def stop(self) -> None:
         self._elapsed_ns = time.perf_counter_ns() - self._t0
         self.end_unix = time.time()
+        if getattr(self, "_nvtx", False):
+            import torch
+
+            torch.cuda.nvtx.range_pop()
+            self._nvtx = False
         stats = TimeStats(
             tags=self.tags,
             elapsed_s=self.elapsed_s,
