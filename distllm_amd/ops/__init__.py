@@ -225,16 +225,25 @@ def linear(x, weight, bias=None, act=ACT_NONE):
                     torch.cuda.synchronize()
                     return _t.perf_counter() - t0
 
-                tb = timeit(lambda: torch.nn.functional.linear(
-                    x2, weight, bias))
-                td = timeit(lambda: _hip_ops.decode_gemm(x2, weight, bias,
-                                                         sk))
-                choice = "decode" if td < tb else "blaslt"
+                times = {
+                    "blaslt": timeit(lambda: torch.nn.functional.linear(
+                        x2, weight, bias)),
+                    "decode": timeit(lambda: _hip_ops.decode_gemm(
+                        x2, weight, bias, sk)),
+                }
+                try:  # hipBLASLt with measured algo search (blt_gemm.cpp)
+                    times["blt_fused"] = timeit(
+                        lambda: _blt_fused_path(x2, weight, bias, act))
+                except RuntimeError:
+                    pass
+                choice = min(times, key=times.get)
                 _GEMM_CHOICE[key] = choice
             if choice == "decode":
                 out = _hip_ops.decode_gemm(x2, weight, bias, sk)
-                return out.reshape(*orig_shape[:-1], N)
-            out = torch.nn.functional.linear(x2, weight, bias)
+            elif choice == "blt_fused":
+                out = _blt_fused_path(x2, weight, bias, act)
+            else:
+                out = torch.nn.functional.linear(x2, weight, bias)
             return out.reshape(*orig_shape[:-1], N)
         shapes_ok = (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
                      and x2.is_contiguous())
