@@ -103,7 +103,16 @@ extern "C" void launch_norm_fwd(const void* x, const void* res, const void* w,
                                 const void* b, void* out, void* sum_out,
                                 int n_rows, int H, float eps, bool rms,
                                 hipStream_t stream) {
-  int blocks = min((n_rows + 3) / 4, 2048);
+  // Grid cap: measured on MI355X (DISTLLM_NORM_BLOCKS sweep in
+  // profiles/RESULTS.md): deep grids beat a 2048-block grid-stride
+  // loop by ~5%; a 2-rows-per-wave ILP variant was also probed and
+  // REGRESSED (149 -> 172 us plain LN) — tail guards in the unrolled
+  // loops cost more than the overlap wins, so R stays 1.
+  static const int cap = [] {
+    const char* e = getenv("DISTLLM_NORM_BLOCKS");
+    return e ? atoi(e) : 32768;
+  }();
+  int blocks = min((n_rows + 3) / 4, cap);
   bool has_res = res != nullptr;
   bool write_sum = sum_out != nullptr;
   int ncl = (H / 8 + WAVE - 1) / WAVE;
