@@ -119,7 +119,15 @@ def generate(
                             "temperature": temperature,
                             "max_tokens": max_tokens}
     output_dir.mkdir(parents=True, exist_ok=True)
-    for f in sorted(Path(input_dir).iterdir()):
+    input_dir = Path(input_dir)
+    if (input_dir / "state.json").exists() or (
+            input_dir / "dataset_info.json").exists():
+        # input_dir IS one saved HF dataset (e.g. `distllm merge` output),
+        # not a directory of shards: process it as a single input.
+        inputs = [input_dir]
+    else:
+        inputs = sorted(input_dir.iterdir())
+    for f in inputs:
         generate_worker(f, output_dir, {"name": prompt_name},
                         {"name": reader_name}, {"name": writer_name},
                         generator_kwargs)

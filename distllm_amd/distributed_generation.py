@@ -63,6 +63,13 @@ def generate_worker(
     text = [text[i] for i in keep]
     results = [results[i] for i in keep]
 
+    if not results:
+        # every response was dropped by postprocess (e.g. none parsed):
+        # write nothing — a 0-row HF dataset dir is not even re-loadable,
+        # and the merge step would have to skip it anyway.
+        print(f"[generate] {input_path}: 0 usable responses, no shard")
+        return
+
     with Timer("wrote-output", str(input_path)):
         from .embed.writers import unique_shard_dir
 

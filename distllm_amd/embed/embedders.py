@@ -115,6 +115,13 @@ class SemanticChunkEmbedder:
         ds = dataloader.dataset
         if ds.metadata is None:
             raise ValueError("Metadata is required for semantic chunking.")
+        if len(ds) == 0:
+            # every buffer was filtered out (all sentences shorter than
+            # the dataset's min_buffer_length): an empty shard, not an
+            # error — one short file must not kill a distributed run.
+            return EmbedderResult(
+                embeddings=np.zeros((0, 0), dtype=np.float32),
+                text=[], metadata=[])
         if ds.metadata[0].get("path") is None:
             raise ValueError("Metadata path is required for semantic "
                              "chunking.")

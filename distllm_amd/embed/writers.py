@@ -56,7 +56,12 @@ class HuggingFaceWriter:
         if not parts:
             raise ValueError("no readable dataset shards to merge")
         merged = concatenate_datasets(parts)
-        merged.save_to_disk(str(output_dir), num_proc=num_proc or None)
+        # num_proc=1 still makes `datasets` spawn a 1-worker pool (which
+        # re-imports __main__ — breaks under embedded interpreters and
+        # wastes a fork); only parallelize for num_proc > 1.
+        merged.save_to_disk(str(output_dir),
+                            num_proc=num_proc if num_proc and num_proc > 1
+                            else None)
 
 
 class NumpyWriterConfig(BaseConfig):

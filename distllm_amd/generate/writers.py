@@ -44,7 +44,8 @@ class HuggingFaceGenWriter:
         if not parts:
             raise ValueError("no readable dataset shards to merge")
         concatenate_datasets(parts).save_to_disk(
-            str(output_dir), num_proc=num_proc or None
+            str(output_dir),
+            num_proc=num_proc if num_proc and num_proc > 1 else None,
         )
 
 

@@ -166,3 +166,30 @@ def test_distributed_embedding_entry(tmp_path):
     shards = list((odir / "embeddings").iterdir())
     assert len(shards) == 2
     assert (odir / "config.yaml").exists()
+
+
+def test_semantic_chunk_all_buffers_filtered(tmp_path):
+    """Docs whose sentences are all shorter than min_buffer_length must
+    produce an EMPTY result, not an IndexError crash — one short file
+    must not kill a distributed embed run."""
+    import json
+
+    from distllm_amd.embed.datasets import (JsonlChunkDataset,
+                                            JsonlChunkDatasetConfig)
+    from distllm_amd.embed.embedders import (SemanticChunkEmbedder,
+                                             SemanticChunkEmbedderConfig)
+    from distllm_amd.embed import get_encoder, get_pooler
+
+    f = tmp_path / "short.jsonl"
+    f.write_text(json.dumps({"text": "Too short. Really.",
+                             "path": "p0"}) + "\n")
+    enc = get_encoder({"name": "auto",
+                       "pretrained_model_name_or_path": "bert-tiny",
+                       "half_precision": False})
+    pooler = get_pooler({"name": "mean"})
+    ds = JsonlChunkDataset(JsonlChunkDatasetConfig())
+    loader = ds.get_dataloader(f, enc)
+    emb = SemanticChunkEmbedder(SemanticChunkEmbedderConfig())
+    result = emb.embed(loader, enc, pooler)
+    assert result.embeddings.shape[0] == 0
+    assert result.text == []
