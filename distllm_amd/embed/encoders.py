@@ -126,8 +126,13 @@ def _build(cfg) -> _NativeEncoder:
     preset = resolve_preset(cfg.pretrained_model_name_or_path)
     dims = get_dims(preset)
     device = _default_device()
-    dtype = torch.bfloat16 if (cfg.half_precision and device != "cpu") \
-        else torch.float32
+    if device != "cpu" and not cfg.half_precision:
+        # The MFMA kernel path is bf16-I/O by design (fp32 accumulate);
+        # fp32 end-to-end is the CPU parity mode. Keep the GPU native
+        # rather than crashing bf16-only kernels with fp32 tensors.
+        print("[encoder] half_precision=False requested on GPU: the "
+              "HIP MFMA path is bf16 (fp32 accumulate); running bf16.")
+    dtype = torch.bfloat16 if device != "cpu" else torch.float32
     tok = load_tokenizer(
         cfg.tokenizer_path or cfg.pretrained_model_name_or_path,
         arch=dims.arch,
