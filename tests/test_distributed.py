@@ -13,7 +13,10 @@ REPO = Path(__file__).parent.parent
 
 
 def _torchrun(args, timeout=300):
-    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               # these are gloo CPU-logic tests: hide any GPUs so
+               # world_size 2 also runs on a 1-GPU box
+               HIP_VISIBLE_DEVICES="", CUDA_VISIBLE_DEVICES="")
     return subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",

@@ -94,7 +94,7 @@ def test_retriever_and_rag_generator(tmp_path):
     tok = enc.tokenizer(texts, return_tensors="pt")
     hidden = enc.encode(tok)
     pooled = pooler.pool(hidden, tok["attention_mask"], normalize=True)
-    emb = pooled.float().numpy()
+    emb = pooled.float().cpu().numpy()
     ds = HFDataset.from_dict({"text": texts, "embeddings": [e for e in emb]})
     d = tmp_path / "c"
     ds.save_to_disk(str(d))
