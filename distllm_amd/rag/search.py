@@ -187,18 +187,37 @@ class HipFlatIndex:
         self.ivf_order = order            # corpus row ids, cluster-major
         self.ivf_offsets = offs           # [C+1] boundaries into order
         self.ivf_nlist = C
+        # cluster-major slab copy: probing then reads CONTIGUOUS rows
+        # (a [len, D] @ [D] GEMV per probed cluster) instead of a random
+        # row gather — 384x less traffic than gathering candidate rows.
+        # Costs one corpus duplicate; by design for 288 GB HBM.
+        self.corpus_sorted = self.corpus[order.to(self.corpus.device)]\
+            .contiguous()
 
     def _topk_ivf(self, q: torch.Tensor, k: int):
+        """IVF probe scoring over cluster-major contiguous slabs.
+
+        IVF is the LOW-LATENCY single/small-query path: it reads only
+        the probed clusters' rows. For large query batches the exact
+        path is faster on this GPU — one GEMM reads the corpus once for
+        ALL queries, while IVF re-reads per query (measured in
+        profiles/RESULTS.md)."""
         nprobe = min(self.nprobe, self.ivf_nlist)
         cscores = q @ self.ivf_centroids.t()
         _, probes = torch.topk(cscores, nprobe, dim=1)   # [Q, nprobe]
+        probes_h = probes.cpu().tolist()
         vs, is_ = [], []
         for qi in range(q.shape[0]):
-            cand = torch.cat([
-                self.ivf_order[self.ivf_offsets[c]:self.ivf_offsets[c + 1]]
-                for c in probes[qi].tolist()
-            ])
-            scores = (self.corpus[cand] @ q[qi]).unsqueeze(0)
+            parts_v, parts_i = [], []
+            for c in probes_h[qi]:
+                o0 = int(self.ivf_offsets[c])
+                o1 = int(self.ivf_offsets[c + 1])
+                if o0 == o1:
+                    continue
+                parts_v.append(self.corpus_sorted[o0:o1] @ q[qi])
+                parts_i.append(self.ivf_order[o0:o1])
+            scores = torch.cat(parts_v).unsqueeze(0)
+            cand = torch.cat(parts_i).to(scores.device)
             kk = min(k, cand.numel())
             v, i = ops.topk(scores.contiguous(), kk)
             idx = cand[i[0].long()].int()
