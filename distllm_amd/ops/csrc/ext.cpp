@@ -48,6 +48,10 @@ void launch_hamming_score(const void*, const void*, float*, int, long, int,
                           hipStream_t);
 void launch_sample(const float*, const float*, int*, int, int, float, float,
                    hipStream_t);
+void launch_paged_attn_split(const void*, const void*, const void*,
+                             const int*, const int*, void*, void*, void*,
+                             int, int, int, int, int, int, float, bool,
+                             hipStream_t);
 void launch_paged_attn(const void*, const void*, const void*, const int*,
                        const int*, void*, int, int, int, int, int, float,
                        bool, hipStream_t);
@@ -299,6 +303,32 @@ torch::Tensor paged_attn(torch::Tensor q, torch::Tensor kcache,
   const int max_blocks = block_table.size(1);
   const bool fp8 = kcache.scalar_type() == at::kByte;
   auto out = torch::empty_like(q);
+  // Flash-decoding context split when B*HKV alone underfills the chip
+  // (small batches): parallelize over context chunks + combine pass.
+  const int GR = (HKV > 0 && H % HKV == 0) ? H / HKV : 0;
+  const bool gr_ok = (GR == 1 || GR == 2 || GR == 4 || GR == 8);
+  int NS = 1;
+  if (gr_ok && B * HKV < 512 && max_blocks > 1)
+    NS = (int)std::min<long>(std::min(512L / (B * HKV) + 1, 16L),
+                             max_blocks);
+  static const int ns_env = [] {  // probe override: 0=auto, 1=off, N
+    const char* e = getenv("DISTLLM_PA_NSPLIT");
+    return e ? atoi(e) : 0;
+  }();
+  if (ns_env > 0 && gr_ok)
+    NS = std::min(ns_env, max_blocks);
+  if (NS > 1) {
+    auto f32 = q.options().dtype(at::kFloat);
+    auto pacc = torch::empty({(long)B * HKV * NS * GR, D}, f32);
+    auto pml = torch::empty({(long)B * HKV * NS * GR, 2}, f32);
+    launch_paged_attn_split(q.data_ptr(), kcache.data_ptr(),
+                            vcache.data_ptr(), block_table.data_ptr<int>(),
+                            ctx_len.data_ptr<int>(), pacc.data_ptr(),
+                            pml.data_ptr(), out.data_ptr(), B, H, HKV, D,
+                            max_blocks, NS, (float)scale, fp8,
+                            cur_stream());
+    return out;
+  }
   launch_paged_attn(q.data_ptr(), kcache.data_ptr(), vcache.data_ptr(),
                     block_table.data_ptr<int>(), ctx_len.data_ptr<int>(),
                     out.data_ptr(), B, H, HKV, D, max_blocks, (float)scale,

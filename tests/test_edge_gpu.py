@@ -101,3 +101,22 @@ def test_paged_attention_short_contexts(device):
     out = ops.paged_attn(q, kc, vc, bt, ctx)
     ref = torch_ref.paged_attn(q, kc, vc, bt, ctx, 1.0 / math.sqrt(D))
     _close(out, ref)
+
+
+def test_paged_attention_long_context_small_batch(device):
+    """B=1 long context exercises the flash-decoding split path
+    (context chunks + combine pass)."""
+    torch.manual_seed(5)
+    B, H, HKV, D, BLK = 1, 32, 8, 128, 16
+    ctx_n = 1999
+    maxb = (ctx_n + BLK - 1) // BLK + 1
+    kc = torch.randn(B * maxb + 1, HKV, BLK, D, device=device,
+                     dtype=torch.bfloat16)
+    vc = torch.randn_like(kc)
+    bt = torch.arange(1, B * maxb + 1, device=device,
+                      dtype=torch.int32).reshape(B, maxb)
+    ctx = torch.tensor([ctx_n], device=device, dtype=torch.int32)
+    q = torch.randn(B, H, D, device=device, dtype=torch.bfloat16)
+    out = ops.paged_attn(q, kc, vc, bt, ctx)
+    ref = torch_ref.paged_attn(q, kc, vc, bt, ctx, 1.0 / math.sqrt(D))
+    _close(out, ref)
