@@ -307,8 +307,12 @@ torch::Tensor paged_attn(torch::Tensor q, torch::Tensor kcache,
   // (small batches): parallelize over context chunks + combine pass.
   const int GR = (HKV > 0 && H % HKV == 0) ? H / HKV : 0;
   const bool gr_ok = (GR == 1 || GR == 2 || GR == 4 || GR == 8);
+  // Measured policy (profiles/RESULTS.md): split pays when B*HKV <= 128
+  // AND the context is long (max_blocks >= 64 i.e. >= 1k tokens) —
+  // B=1 ctx8192: 322 -> 50 us; B=32 ctx2048 regressed with split, so
+  // larger batches keep the one-pass kernel.
   int NS = 1;
-  if (gr_ok && B * HKV < 512 && max_blocks > 1)
+  if (gr_ok && B * HKV <= 128 && max_blocks >= 64)
     NS = (int)std::min<long>(std::min(512L / (B * HKV) + 1, 16L),
                              max_blocks);
   static const int ns_env = [] {  // probe override: 0=auto, 1=off, N
