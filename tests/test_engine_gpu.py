@@ -72,3 +72,35 @@ def test_fp8_kv_cache_decode(model):
     agree = sum(a == b for x, y in zip(o16, o8) for a, b in zip(x, y))
     total = sum(len(x) for x in o16)
     assert agree / total > 0.6, f"fp8 KV diverged: {o8} vs {o16}"
+
+
+def test_engine_loop_stress_gpu(model):
+    """Continuous-batching churn: 100 concurrent requests with mixed
+    prompt lengths and token budgets through the background EngineLoop;
+    every request must complete with the right token count and valid
+    ids (paged-KV alloc/free under admission pressure)."""
+    import random
+    from concurrent.futures import ThreadPoolExecutor
+
+    from distllm_amd.engine import SamplingParams
+    from distllm_amd.engine.loop import EngineLoop
+
+    rng = random.Random(0)
+    engine = LLMEngine(model, max_batch_size=16, max_model_len=512,
+                       num_blocks=2048)
+    params = SamplingParams(temperature=0.7, min_p=0.05, max_tokens=16,
+                            stop_token_ids=())
+    loop = EngineLoop(engine, params)
+    reqs = [([rng.randrange(4, model.dims.vocab_size)
+              for _ in range(rng.randrange(1, 200))],
+             rng.randrange(1, 24)) for _ in range(100)]
+    try:
+        with ThreadPoolExecutor(max_workers=32) as ex:
+            futs = [ex.submit(loop.generate, [p], mt) for p, mt in reqs]
+            outs = [f.result(timeout=300)[0] for f in futs]
+    finally:
+        loop.close()
+    assert len(outs) == 100
+    for (p, mt), o in zip(reqs, outs):
+        assert len(o) == mt, f"expected {mt} tokens, got {len(o)}"
+        assert all(0 <= t < model.dims.vocab_size for t in o)
