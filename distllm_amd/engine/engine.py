@@ -44,6 +44,8 @@ class _Seq:
     out_ids: list[int] = field(default_factory=list)
     blocks: list[int] = field(default_factory=list)
     done: bool = False
+    # per-sequence token budget; None -> the batch SamplingParams cap
+    max_tokens: int | None = None
 
     @property
     def length(self) -> int:
@@ -271,7 +273,7 @@ class LLMEngine:
             s.out_ids.append(int(tok))
             stop = (
                 tok in params.stop_token_ids
-                or len(s.out_ids) >= params.max_tokens
+                or len(s.out_ids) >= (s.max_tokens or params.max_tokens)
                 or s.length >= self.max_model_len
             )
             if stop:

@@ -78,7 +78,8 @@ class EngineLoop:
                     break
                 self._waiting.pop(0)
                 seq = _Seq(seq_id=id(req),
-                           prompt_ids=req.prompt_ids[-plen:])
+                           prompt_ids=req.prompt_ids[-plen:],
+                           max_tokens=req.max_tokens)
                 seq.blocks = eng.blockman.alloc(need)
                 req.seq = seq
                 admitted.append(req)
@@ -99,8 +100,6 @@ class EngineLoop:
             if admitted:
                 try:
                     seqs = [r.seq for r in admitted]
-                    # per-request max_tokens: cap with the smallest; the
-                    # stop check below enforces each request's own limit
                     eng._run_prefill(seqs, params, results)
                 except Exception as e:
                     for r in admitted:
