@@ -28,6 +28,11 @@ class HuggingFaceWriter:
     def write(self, output_dir: Path, result: EmbedderResult) -> None:
         from datasets import Dataset as HFDataset
 
+        if len(result.text) == 0:
+            # a 0-row saved HF dataset cannot be load_from_disk'd back;
+            # write nothing (merge would have to skip it anyway)
+            print(f"[write] {output_dir}: empty result, no shard written")
+            return
         rows: dict[str, list] = {
             "text": result.text,
             "embeddings": [e for e in result.embeddings],
