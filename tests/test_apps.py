@@ -207,3 +207,23 @@ def test_prompt_postprocess_behaviors():
 
     post = amp.postprocess(["Q: which peptide?"])
     assert _json.loads(post[0])["question"] == "Q: which peptide?"
+
+
+def test_generate_worker_skips_empty_results(tmp_path, monkeypatch):
+    """All responses dropped by postprocess -> no unloadable 0-row
+    shard is written."""
+    from datasets import Dataset as HFDataset
+
+    from distllm_amd.distributed_generation import generate_worker
+
+    HFDataset.from_dict({"text": ["short chunk"],
+                         "path": ["p0"]}).save_to_disk(str(tmp_path / "in"))
+    out = tmp_path / "gen"
+    out.mkdir()
+    # question_chunk postprocess drops answers without parseable output;
+    # llama-tiny random weights produce none at max_tokens=2
+    generate_worker(tmp_path / "in", out, {"name": "question_chunk"},
+                    {"name": "huggingface"}, {"name": "huggingface"},
+                    {"name": "native", "model": "llama-tiny",
+                     "max_tokens": 2, "max_batch_size": 2})
+    assert list(out.iterdir()) == []

@@ -193,3 +193,18 @@ def test_semantic_chunk_all_buffers_filtered(tmp_path):
     result = emb.embed(loader, enc, pooler)
     assert result.embeddings.shape[0] == 0
     assert result.text == []
+
+
+def test_hf_writer_skips_empty_result(tmp_path):
+    import numpy as np
+
+    from distllm_amd.embed.embedders import EmbedderResult
+    from distllm_amd.embed.writers import get_writer
+
+    w = get_writer({"name": "huggingface"})
+    out = tmp_path / "shard"
+    w.write(out, EmbedderResult(
+        embeddings=np.zeros((0, 0), dtype=np.float32), text=[],
+        metadata=[]))
+    # nothing written: an empty saved HF dataset cannot be loaded back
+    assert not out.exists() or not any(out.iterdir())
