@@ -149,6 +149,13 @@ def load_weights(model, path: str | Path) -> None:
             )
     for name, p in own.items():
         if name in mapped:
-            p.copy_(mapped[name].to(p.dtype))
+            w = mapped[name]
+            if tuple(w.shape) != tuple(p.shape):
+                raise ValueError(
+                    f"checkpoint weight {name} has shape "
+                    f"{tuple(w.shape)}, model expects {tuple(p.shape)} — "
+                    f"wrong preset for this checkpoint?"
+                )
+            p.copy_(w.to(p.dtype))
         else:
             print(f"[loader] warning: no weight for {name}, keeping init")
