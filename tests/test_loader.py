@@ -123,3 +123,45 @@ def test_loader_shape_mismatch_message(tmp_path):
     dst = build_model("bert-tiny", device="cpu", dtype=torch.float32)
     with pytest.raises(ValueError, match="wrong preset"):
         load_weights(dst, tmp_path / "ck")
+
+
+def _esm2_to_hf(model) -> dict:
+    p = dict(model.named_parameters())
+    H = model.dims.hidden_size
+    out = {
+        "esm.embeddings.word_embeddings.weight": p["word_emb"],
+        "esm.encoder.emb_layer_norm_after.weight": p["final_ln_w"],
+        "esm.encoder.emb_layer_norm_after.bias": p["final_ln_b"],
+    }
+    for i in range(len(model.layers)):
+        lp = f"esm.encoder.layer.{i}."
+        q, k, v = p[f"layers.{i}.wqkv"].split(H, dim=0)
+        qb, kb, vb = p[f"layers.{i}.bqkv"].split(H, dim=0)
+        out[lp + "attention.self.query.weight"] = q
+        out[lp + "attention.self.key.weight"] = k
+        out[lp + "attention.self.value.weight"] = v
+        out[lp + "attention.self.query.bias"] = qb
+        out[lp + "attention.self.key.bias"] = kb
+        out[lp + "attention.self.value.bias"] = vb
+        out[lp + "attention.output.dense.weight"] = p[f"layers.{i}.wo"]
+        out[lp + "attention.output.dense.bias"] = p[f"layers.{i}.bo"]
+        out[lp + "attention.LayerNorm.weight"] = p[f"layers.{i}.ln1_w"]
+        out[lp + "attention.LayerNorm.bias"] = p[f"layers.{i}.ln1_b"]
+        out[lp + "LayerNorm.weight"] = p[f"layers.{i}.ln2_w"]
+        out[lp + "LayerNorm.bias"] = p[f"layers.{i}.ln2_b"]
+        out[lp + "intermediate.dense.weight"] = p[f"layers.{i}.w1"]
+        out[lp + "intermediate.dense.bias"] = p[f"layers.{i}.b1"]
+        out[lp + "output.dense.weight"] = p[f"layers.{i}.w2"]
+        out[lp + "output.dense.bias"] = p[f"layers.{i}.b2"]
+    return out
+
+
+def test_esm2_hf_mapping_roundtrip(tmp_path):
+    torch.manual_seed(7)
+    src = build_model("esm2-8m", device="cpu", dtype=torch.float32)
+    _save(tmp_path / "ck", _esm2_to_hf(src))
+    dst = build_model("esm2-8m", device="cpu", dtype=torch.float32)
+    load_weights(dst, tmp_path / "ck")
+    ids = torch.randint(4, src.dims.vocab_size, (2, 10))
+    mask = torch.ones_like(ids, dtype=torch.int32)
+    assert torch.allclose(src(ids, mask), dst(ids, mask), atol=1e-5)
