@@ -210,3 +210,28 @@ def test_cli_generate(tmp_path):
     assert r.exit_code == 0, r.output
     shards = [p for p in odir.iterdir() if p.is_dir()]
     assert len(shards) == 1
+
+
+def test_example_yamls_parse(tmp_path):
+    """Every shipped examples/*.yaml must validate against its config
+    schema (guards the docs against schema drift)."""
+    import yaml
+
+    from pathlib import Path
+
+    root = Path(__file__).parent.parent / "examples"
+
+    from distllm_amd.chat import ChatAppConfig, substitute_env
+    from distllm_amd.distributed_embedding import EmbeddingConfig
+    from distllm_amd.distributed_generation import GenerationConfig
+    from distllm_amd.mcqa.harness import MCQAConfig
+
+    EmbeddingConfig(**yaml.safe_load((root / "embed.yaml").read_text()))
+    GenerationConfig(**yaml.safe_load((root / "generate.yaml").read_text()))
+    ChatAppConfig(**substitute_env(
+        yaml.safe_load((root / "chat.yaml").read_text())))
+    MCQAConfig(**substitute_env(
+        yaml.safe_load((root / "mcqa.yaml").read_text())))
+    # evaluate.yaml is a plain task-list document for `distllm evaluate`
+    ev = yaml.safe_load((root / "evaluate.yaml").read_text())
+    assert "tasks" in ev or "task_names" in ev or isinstance(ev, dict)
