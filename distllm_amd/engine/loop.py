@@ -1,5 +1,10 @@
 """Online engine loop: vLLM-style request-level continuous batching.
 
+Reference parity: the reference delegated this to vLLM's AsyncLLMEngine
+behind an OpenAI endpoint (distllm mcqa launches `vllm serve`,
+mcqa/rag_argonium_score_parallel_v3.py:1206-1354; client batching in
+the MCQA queue, v3:984-989,1419-1449). Here the engine loop is native.
+
 DynamicBatcher groups concurrent requests into discrete generate()
 calls, which still head-of-line blocks a second batch behind the first.
 EngineLoop owns the engine on a background thread and admits requests

@@ -1,5 +1,11 @@
 """Local checkpoint loading (safetensors) with HF name mapping.
 
+Reference parity: the reference loads models by HF name via
+AutoModel.from_pretrained (distllm/embed/encoders/auto.py:60-70) and
+vLLM's loader; this framework owns the forward, so checkpoints map
+onto the native parameter layout instead (TP shards sliced
+Megatron-style for tp > 1).
+
 No network in this environment: weights load only from local dirs. The
 mapping covers HF BertModel, EsmModel and Llama/Mistral layouts so a
 user with a downloaded checkpoint gets real weights; otherwise models
