@@ -70,8 +70,10 @@ def barrier() -> None:
 
 def shard_list(items: Sequence[T], rank: int | None = None,
                world: int | None = None) -> list[T]:
-    """Contiguous shard of the input list for this rank (file-level data
-    parallelism; replaces ParslPoolExecutor.map fan-out)."""
+    """Strided (round-robin) shard of the input list for this rank —
+    balances sorted-by-name inputs whose sizes correlate with position
+    (file-level data parallelism; replaces ParslPoolExecutor.map
+    fan-out)."""
     r = get_rank() if rank is None else rank
     w = get_world_size() if world is None else world
     return list(items[r::w])
