@@ -30,6 +30,10 @@ class NativeGeneratorConfig(BaseConfig):
     temperature: float = 0.5
     min_p: float = 0.1
     top_p: float = 0.0
+    # num_beams > 1 switches to beam search (the reference's
+    # huggingface backend semantics, huggingface_backend.py:41-44);
+    # 0/1 = the sampling engine path
+    num_beams: int = 0
     max_tokens: int = 2000
     max_batch_size: int = 64
     max_model_len: int = 4096
@@ -126,7 +130,17 @@ class NativeGenerator:
         if isinstance(prompts, str):
             prompts = [prompts]
         prompt_ids = self._tokenize(prompts)
-        if self._loop is not None:  # loop owns the engine once started
+        if self.config.num_beams and self.config.num_beams > 1:
+            from ..engine.beam import beam_generate
+
+            eos = getattr(self.tokenizer, "eos_token_id", None)
+            out_ids = [
+                beam_generate(self.engine.model, p,
+                              self.config.num_beams,
+                              self.params.max_tokens, eos_token_id=eos)
+                for p in prompt_ids
+            ]
+        elif self._loop is not None:  # loop owns the engine once started
             out_ids = self._loop.generate(prompt_ids)
         else:
             out_ids = self.engine.generate(prompt_ids, self.params)
