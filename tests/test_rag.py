@@ -148,3 +148,35 @@ def test_ivf_search(tmp_path):
         for a, b in zip(res.total_indices, ref.total_indices)
     ) / 15.0
     assert overlap >= 0.8, overlap
+
+
+def test_run_eval_suite_end_to_end(tmp_path):
+    """Full eval-suite run offline: fixture task data pre-placed (skips
+    curl), native tiny generator, metrics written to the results JSON
+    (reference rag/evaluate.py:18-99 + tasks)."""
+    import json
+
+    from distllm_amd.rag.evaluate import EvalSuiteConfig, run_eval_suite
+
+    dl = tmp_path / "eval_data"
+    (dl / "sciq").mkdir(parents=True)
+    (dl / "sciq" / "test.json").write_text(json.dumps([{
+        "question": "What is water",
+        "correct_answer": "H2O",
+        "distractor1": "CO2", "distractor2": "O2", "distractor3": "N2",
+    }] * 3))
+    cfg = EvalSuiteConfig(
+        rag_configs=[{
+            "generator_config": {"name": "native", "model": "llama-tiny",
+                                 "max_tokens": 4, "max_batch_size": 4,
+                                 "max_model_len": 256},
+        }],
+        tasks=["sciq"],
+        download_dir=dl,
+        output_path=tmp_path / "results.json",
+    )
+    rows = run_eval_suite(cfg)
+    assert len(rows) == 1
+    assert rows[0]["task"] == "sciq" and 0.0 <= rows[0]["accuracy"] <= 1.0
+    saved = json.loads((tmp_path / "results.json").read_text())
+    assert saved == rows
