@@ -121,3 +121,99 @@ def test_ref_topk_vs_sort():
     assert torch.allclose(v, want_v[:, :7])
     got = scores.gather(1, i.long())
     assert torch.allclose(got, v)
+
+
+def test_ref_gelu_vs_closed_form():
+    x = torch.linspace(-4, 4, 33)
+    out_erf = torch_ref.bias_gelu(x, None, erf=True)
+    want = x * 0.5 * (1 + torch.erf(x / math.sqrt(2)))
+    assert torch.allclose(out_erf, want, atol=1e-6)
+    out_tanh = torch_ref.bias_gelu(x, None, erf=False)
+    inner = math.sqrt(2 / math.pi) * (x + 0.044715 * x ** 3)
+    want_t = 0.5 * x * (1 + torch.tanh(inner))
+    assert torch.allclose(out_tanh, want_t, atol=1e-5)
+    # swiglu = silu(gate) * up
+    g, u = torch.randn(4, 8), torch.randn(4, 8)
+    assert torch.allclose(torch_ref.swiglu(g, u),
+                          torch.nn.functional.silu(g) * u, atol=1e-6)
+
+
+def test_ref_rope_rotation_properties():
+    """RoPE must preserve norms and make QK^T depend only on relative
+    position."""
+    torch.manual_seed(6)
+    D = 16
+    cs, sn = torch_ref.rope_tables(64, D)
+    x = torch.randn(1, 1, 1, D)
+    for p in (0, 5, 31):
+        r = torch_ref.rope_apply(x, cs, sn, torch.tensor([p]))
+        assert torch.allclose(r.norm(), x.norm(), atol=1e-5)
+    q = torch.randn(1, 1, 1, D)
+    k = torch.randn(1, 1, 1, D)
+
+    def dot(pq, pk):
+        rq = torch_ref.rope_apply(q, cs, sn, torch.tensor([pq]))
+        rk = torch_ref.rope_apply(k, cs, sn, torch.tensor([pk]))
+        return float(rq.flatten() @ rk.flatten())
+
+    assert abs(dot(3, 1) - dot(10, 8)) < 1e-4   # same offset
+    assert abs(dot(3, 1) - dot(4, 1)) > 1e-6    # different offset
+
+
+def test_ref_hamming_vs_python_popcount():
+    torch.manual_seed(7)
+    Q, W, N = 3, 2, 5
+    qc = torch.randint(-(2 ** 62), 2 ** 62, (Q, W), dtype=torch.int64)
+    ct = torch.randint(-(2 ** 62), 2 ** 62, (W, N), dtype=torch.int64)
+    out = torch_ref.hamming_score(qc, ct)
+    for qi in range(Q):
+        for n in range(N):
+            ham = sum(
+                bin((int(qc[qi, w]) ^ int(ct[w, n])) & (2 ** 64 - 1))
+                .count("1") for w in range(W))
+            assert out[qi, n] == -ham
+
+
+def test_ref_sampling_distribution():
+    """Temperature sampling via inverse-CDF must hit each token with
+    ~softmax probability; min_p masks low-probability tokens."""
+    torch.manual_seed(8)
+    logits = torch.tensor([[2.0, 1.0, -1.0, -10.0]]).repeat(4000, 1)
+    rand = torch.rand(4000)
+    ids = torch_ref.sample_tokens(logits, rand, temp=1.0, min_p=0.0)
+    freq = torch.bincount(ids.long(), minlength=4).float() / 4000
+    want = torch.softmax(logits[0], -1)
+    assert (freq - want).abs().max() < 0.03
+    # min_p=0.2: tokens below 0.2*max_prob are never sampled
+    ids2 = torch_ref.sample_tokens(logits, rand, temp=1.0, min_p=0.2)
+    assert set(ids2.unique().tolist()) <= {0, 1}
+    # temp=0 is argmax
+    ids3 = torch_ref.sample_tokens(logits, rand, temp=0.0, min_p=0.5)
+    assert (ids3 == 0).all()
+
+
+def test_ref_l2_and_adjacent_cosine():
+    torch.manual_seed(9)
+    x = torch.randn(5, 8)
+    n = torch_ref.l2_normalize(x)
+    assert torch.allclose(n.norm(dim=1), torch.ones(5), atol=1e-5)
+    d = torch_ref.adjacent_cosine(n)
+    for i in range(4):
+        cos = float(n[i] @ n[i + 1])
+        assert abs(float(d[i]) - (1 - cos)) < 1e-5
+
+
+def test_ref_last_token_pool_vs_naive():
+    torch.manual_seed(10)
+    h = torch.randn(2, 6, 8)
+    mask = torch.tensor([[1, 1, 1, 0, 0, 0], [1, 1, 1, 1, 1, 1]],
+                        dtype=torch.int32)
+    out = torch_ref.last_token_pool(h, mask, left_padded=False,
+                                    normalize=False)
+    assert torch.allclose(out[0], h[0, 2])
+    assert torch.allclose(out[1], h[1, 5])
+    # left padding: last token is always the final position
+    mask_l = torch.tensor([[0, 0, 0, 1, 1, 1]], dtype=torch.int32)
+    out_l = torch_ref.last_token_pool(h[:1], mask_l, left_padded=True,
+                                      normalize=False)
+    assert torch.allclose(out_l[0], h[0, 5])
