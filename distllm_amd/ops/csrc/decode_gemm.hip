@@ -1,4 +1,8 @@
 // Skinny-M (decode-batch) split-K MFMA GEMM: out[M,N] = x[M,K] @ W[N,K]^T.
+// Replaces the skinny-M GEMMs vLLM issues per decode step (reference
+// delegated generation to vLLM, distllm/generate/generators/
+// vllm_backend.py:30-68): M <= 256 rows against
+// large [N, K] weights is too narrow to fill 256 CUs without split-K.
 //
 // hipBLASLt runs Llama-decode projections at 20-70% of the
 // weight-streaming roofline (profiles/decode_gemm_probe.py); this

@@ -1,4 +1,7 @@
 // Torch bindings for the distllm-amd HIP kernel library (gfx950).
+// Python bindings for every native op (SURVEY.md §2.9 K1-K17 kernel
+// surface). The ops layer (distllm_amd/ops/__init__.py) fails loudly
+// on GPU if this extension is missing — no silent eager fallback.
 // All kernels are hand-written CDNA4 HIP (see *.hip in this directory);
 // this file only validates tensors and forwards device pointers.
 #include <torch/extension.h>

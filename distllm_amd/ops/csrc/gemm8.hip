@@ -1,4 +1,8 @@
 // Deep-pipelined CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T (+bias/act).
+// Replaces the dense projection GEMMs inside the reference's HF/vLLM
+// model forwards (reference distllm/embed/encoders/auto.py encode();
+// SURVEY.md §2.9 K1/K2 rows) for shapes where it beats hipBLASLt —
+// the per-shape winner is picked by runtime autotune in ops.linear.
 //
 // 256x256x64 tile, 512 threads = 8 waves, each wave computes ALL 256
 // rows x 32 cols (16x2 fragments of mfma_f32_16x16x32_bf16). This
