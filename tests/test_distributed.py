@@ -12,15 +12,24 @@ import yaml
 REPO = Path(__file__).parent.parent
 
 
+def _free_port() -> int:
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _torchrun(args, timeout=300):
     env = dict(os.environ, MASTER_ADDR="127.0.0.1",
                # these are gloo CPU-logic tests: hide any GPUs so
                # world_size 2 also runs on a 1-GPU box
                HIP_VISIBLE_DEVICES="", CUDA_VISIBLE_DEVICES="")
+    # fresh port per call: a fixed port flakes on TIME_WAIT reuse
     return subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29781", *args],
+         "--master-port", str(_free_port()), *args],
         capture_output=True, text=True, timeout=timeout, env=env,
         cwd=REPO,
     )
