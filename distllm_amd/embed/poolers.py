@@ -1,6 +1,12 @@
 """Pooler strategies: mean (excl. BOS/EOS) and last-token, backed by the
 fused HIP pooling kernels (pooling.hip). Reference parity:
 distllm/embed/poolers/{mean,last_token}.py.
+
+Deliberate divergence: the reference mean pooler's
+`attention_mask[:, seq_lengths-1] = 0` vector-index zeroes the column at
+EVERY row's length-1 across all rows of a mixed-length batch; we exclude
+each row's own last real token only (the intended semantics; identical
+for uniform-length batches). See ops/torch_ref.py:mean_pool.
 """
 
 from __future__ import annotations

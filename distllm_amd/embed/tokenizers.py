@@ -134,9 +134,19 @@ def _pad_rows(rows, pad_id, padding, pad_to_multiple_of):
 
 
 def load_tokenizer(name_or_path: str, arch: str = "bert",
-                   vocab_size: int = 30528, model_max_length: int = 512):
+                   vocab_size: int = 30528, model_max_length: int = 512,
+                   allow_hash_fallback: bool | None = None):
     """Local HF tokenizer if files exist; native ESM alphabet for protein
-    archs; deterministic hash tokenizer otherwise."""
+    archs; deterministic hash tokenizer otherwise.
+
+    The hash fallback produces meaningless embeddings on real text — it
+    exists for synthetic benchmarks and random-weight tests only, so it
+    requires explicit opt-in (``allow_hash_fallback=True`` or env
+    ``DISTLLM_ALLOW_HASH_TOKENIZER=1``) and emits a loud warning either
+    way. Without opt-in, an unrecognized model string raises."""
+    import os
+    import warnings
+
     p = Path(name_or_path) if name_or_path else None
     if p is not None and p.exists():
         from transformers import AutoTokenizer
@@ -145,5 +155,24 @@ def load_tokenizer(name_or_path: str, arch: str = "bert",
         return tok
     if arch in ("esm2", "esmc"):
         return EsmTokenizer(model_max_length=model_max_length)
+    if allow_hash_fallback is None:
+        allow_hash_fallback = os.environ.get(
+            "DISTLLM_ALLOW_HASH_TOKENIZER", ""
+        ) not in ("", "0")
+    if not allow_hash_fallback:
+        raise ValueError(
+            f"no local tokenizer found at {name_or_path!r} and the "
+            "HashWordTokenizer fallback is not enabled. On real text "
+            "corpora the hash tokenizer yields meaningless embeddings; "
+            "point tokenizer_path at a local HF tokenizer directory, or "
+            "opt in for synthetic/bench use with "
+            "DISTLLM_ALLOW_HASH_TOKENIZER=1 (or allow_hash_fallback=True)."
+        )
+    warnings.warn(
+        f"[tokenizer] no local tokenizer at {name_or_path!r}: using the "
+        "deterministic HashWordTokenizer (synthetic-bench mode). "
+        "Embeddings of real text will be meaningless.",
+        UserWarning, stacklevel=2,
+    )
     return HashWordTokenizer(vocab_size=vocab_size,
                              model_max_length=model_max_length)

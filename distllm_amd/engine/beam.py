@@ -21,6 +21,12 @@ def beam_generate(model, prompt_ids: list[int], num_beams: int,
     """Return the generated ids (prompt excluded) of the best beam by
     length-normalized cumulative log-probability."""
     dev = model.device
+    # Clamp to the model's positional range (the sampling engine path
+    # truncates prompts too): long prompts must not index RoPE tables
+    # out of range mid-search.
+    max_pos = int(model.dims.max_position_embeddings)
+    prompt_ids = list(prompt_ids)[-(max_pos - 1):]
+    max_tokens = min(max_tokens, max_pos - len(prompt_ids))
     beams: list[tuple[list[int], float]] = [(list(prompt_ids), 0.0)]
     done: list[tuple[list[int], float]] = []
 

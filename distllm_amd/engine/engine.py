@@ -296,12 +296,16 @@ class LLMEngine:
                       params: SamplingParams) -> torch.Tensor:
         t = max(params.temperature, 1e-5)
         probs = torch.softmax(logits.float() / t, dim=-1)
+        if params.min_p > 0.0:
+            floor = params.min_p * probs.max(dim=-1, keepdim=True).values
+            probs = probs.masked_fill(probs < floor, 0.0)
         sorted_p, sorted_i = probs.sort(dim=-1, descending=True)
         cum = sorted_p.cumsum(-1)
         cut = cum - sorted_p > params.top_p
         sorted_p = sorted_p.masked_fill(cut, 0.0)
         sorted_p = sorted_p / sorted_p.sum(-1, keepdim=True)
-        choice = torch.multinomial(sorted_p, 1)
+        # seeded engine RNG: keeps sampling identical across TP ranks
+        choice = torch.multinomial(sorted_p, 1, generator=self._rng)
         return sorted_i.gather(1, choice).squeeze(1).int()
 
 

@@ -13,7 +13,7 @@ from __future__ import annotations
 import os
 from typing import Any, Literal, Union
 
-from pydantic import Field
+from pydantic import Field, model_validator
 
 from ..registry import registry
 from ..utils import BaseConfig, retry
@@ -23,6 +23,18 @@ class NativeGeneratorConfig(BaseConfig):
     """In-process generation on the native HIP engine."""
 
     name: Literal["native", "vllm", "huggingface"] = "native"
+
+    @model_validator(mode="before")
+    @classmethod
+    def _huggingface_defaults(cls, data: Any) -> Any:
+        # Reference HuggingFaceGenerator defaults differ from vLLM's:
+        # num_beams=10, top_p=.95, do_sample=True
+        # (huggingface_backend.py:41-44). YAMLs naming the huggingface
+        # backend and relying on defaults get those semantics here too.
+        if isinstance(data, dict) and data.get("name") == "huggingface":
+            data.setdefault("num_beams", 10)
+            data.setdefault("top_p", 0.95)
+        return data
     model: str = "llama-tiny"
     weights_path: str = ""
     tokenizer_path: str = ""

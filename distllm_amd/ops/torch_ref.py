@@ -63,7 +63,14 @@ def bias_add(x: torch.Tensor, bias: torch.Tensor):
 def mean_pool(hidden: torch.Tensor, mask: torch.Tensor, normalize: bool):
     """Masked mean excluding BOS (pos 0) and the last real token (pos L-1).
 
-    Semantics of reference distllm/embed/poolers/mean.py:13-49.
+    Role of reference distllm/embed/poolers/mean.py:13-49, with ONE
+    deliberate divergence: the reference's `attention_mask[:, L-1] = 0`
+    uses a VECTOR column index, so for a mixed-length batch it zeroes
+    the column at every row's length-1 across ALL rows (excluding
+    interior tokens of longer rows). We exclude each row's OWN last
+    real token only — the per-row behavior the reference intends
+    ("Set the attention mask to 0 for start and end tokens"). Equal
+    results for uniform-length batches.
     hidden [R, S, H]; mask [R, S] int -> [R, H] fp32.
     """
     hf = hidden.float()

@@ -65,7 +65,11 @@ class HipIndexConfig(BaseConfig):
     dataset_chunk_paths: list[Path] | None = None
     precision: str = "float32"
     search_algorithm: str = "exact"
-    rescore_multiplier: int = 2
+    # The reference default is 2 (sentence-transformers semantics) but
+    # sign-bit quantization recall is strongly multiplier-dependent
+    # (recall@10 ~0.2 at x2 on clustered data, profiles/RESULTS.md);
+    # default raised to 8 and a recall warning is emitted for ubinary.
+    rescore_multiplier: int = 8
     num_quantization_workers: int = 1
     nlist: int = 0    # ivf: clusters (0 -> ~sqrt(N))
     nprobe: int = 8   # ivf: clusters searched per query
@@ -79,13 +83,23 @@ class HipFlatIndex:
                  dataset_chunk_paths: list[Path] | None = None,
                  precision: str = "float32",
                  search_algorithm: str = "exact",
-                 rescore_multiplier: int = 2,
+                 rescore_multiplier: int = 8,
                  num_quantization_workers: int = 1,
                  device: str | None = None,
                  score_chunk_rows: int = 1 << 21,
                  nlist: int = 0, nprobe: int = 8):
         if precision not in ("float32", "ubinary"):
             raise ValueError(f"invalid precision {precision}")
+        if precision == "ubinary":
+            import warnings
+
+            warnings.warn(
+                "[hip_index] ubinary (packed sign-bit) precision trades "
+                "recall for a 32x smaller index; recall@k depends "
+                "strongly on the data and on rescore_multiplier "
+                f"(currently {rescore_multiplier}). Measure recall "
+                "against precision='float32' on a sample before "
+                "production use.", UserWarning, stacklevel=2)
         if search_algorithm not in ("exact", "hnsw", "ivf"):
             raise ValueError(f"invalid search_algorithm {search_algorithm}")
         if search_algorithm == "hnsw":
@@ -288,7 +302,15 @@ class HipFlatIndex:
         v = v.cpu().tolist()
         i = i.cpu().tolist()
         print(f"Search time: {time.perf_counter() - t0:.6f} seconds")
-        results = BatchedSearchResults(total_scores=v, total_indices=i)
+        # IVF pads short candidate lists with index -1 / score -inf;
+        # drop those unconditionally (not only under score_threshold) so
+        # -1 never reaches get()/dataset indexing as "last row".
+        vv, ii = [], []
+        for scores, indices in zip(v, i):
+            keep = [(s, ix) for s, ix in zip(scores, indices) if ix >= 0]
+            vv.append([s for s, _ in keep])
+            ii.append([ix for _, ix in keep])
+        results = BatchedSearchResults(total_scores=vv, total_indices=ii)
         return self._filter_search_by_score(results, score_threshold)
 
     def _filter_search_by_score(self, results: BatchedSearchResults,
