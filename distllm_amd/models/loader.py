@@ -105,12 +105,22 @@ def _map_llama(src: dict, model) -> dict:
     return out
 
 
+def _ln(src: dict, prefix: str, which: str) -> torch.Tensor:
+    """LayerNorm param under either modern (weight/bias) or the legacy
+    gamma/beta spelling transformers re-emits for ESM on save."""
+    legacy = {"weight": "gamma", "bias": "beta"}[which]
+    if f"{prefix}.{which}" in src:
+        return src[f"{prefix}.{which}"]
+    return src[f"{prefix}.{legacy}"]
+
+
 def _map_esm2(src: dict, model) -> dict:
     out = {}
     p = "esm." if any(k.startswith("esm.") for k in src) else ""
     out["word_emb"] = src[f"{p}embeddings.word_embeddings.weight"]
-    out["final_ln_w"] = src[f"{p}encoder.emb_layer_norm_after.weight"]
-    out["final_ln_b"] = src[f"{p}encoder.emb_layer_norm_after.bias"]
+    out["final_ln_w"] = _ln(src, f"{p}encoder.emb_layer_norm_after",
+                            "weight")
+    out["final_ln_b"] = _ln(src, f"{p}encoder.emb_layer_norm_after", "bias")
     for i in range(len(model.layers)):
         lp = f"{p}encoder.layer.{i}."
         out[f"layers.{i}.wqkv"] = torch.cat(
@@ -123,10 +133,12 @@ def _map_esm2(src: dict, model) -> dict:
              src[lp + "attention.self.value.bias"]], 0)
         out[f"layers.{i}.wo"] = src[lp + "attention.output.dense.weight"]
         out[f"layers.{i}.bo"] = src[lp + "attention.output.dense.bias"]
-        out[f"layers.{i}.ln1_w"] = src[lp + "attention.LayerNorm.weight"]
-        out[f"layers.{i}.ln1_b"] = src[lp + "attention.LayerNorm.bias"]
-        out[f"layers.{i}.ln2_w"] = src[lp + "LayerNorm.weight"]
-        out[f"layers.{i}.ln2_b"] = src[lp + "LayerNorm.bias"]
+        out[f"layers.{i}.ln1_w"] = _ln(src, lp + "attention.LayerNorm",
+                                       "weight")
+        out[f"layers.{i}.ln1_b"] = _ln(src, lp + "attention.LayerNorm",
+                                       "bias")
+        out[f"layers.{i}.ln2_w"] = _ln(src, lp + "LayerNorm", "weight")
+        out[f"layers.{i}.ln2_b"] = _ln(src, lp + "LayerNorm", "bias")
         out[f"layers.{i}.w1"] = src[lp + "intermediate.dense.weight"]
         out[f"layers.{i}.b1"] = src[lp + "intermediate.dense.bias"]
         out[f"layers.{i}.w2"] = src[lp + "output.dense.weight"]
