@@ -177,6 +177,9 @@ def _autotune_gemm(x2, weight, bias, act):
         "gemm8": timeit(lambda: _hip_ops.gemm8_bf16(x2, weight, bias, act)),
         "blt_fused": timeit(lambda: _blt_fused_path(x2, weight, bias, act)),
     }
+    if x2.shape[1] % 128 == 0:  # gemm16 needs K % 128 == 0
+        times["gemm16"] = timeit(
+            lambda: _hip_ops.gemm16_bf16(x2, weight, bias, act))
     return min(times, key=times.get)
 
 
@@ -250,6 +253,8 @@ def linear(x, weight, bias=None, act=ACT_NONE):
         if shapes_ok and mode != "blaslt":
             if mode == "gemm8":
                 out = _hip_ops.gemm8_bf16(x2, weight, bias, act)
+            elif mode == "gemm16":
+                out = _hip_ops.gemm16_bf16(x2, weight, bias, act)
             else:  # auto
                 key = (min(M, 1 << 20) // 4096, N, K, act)
                 choice = _GEMM_CHOICE.get(key)
@@ -258,6 +263,8 @@ def linear(x, weight, bias=None, act=ACT_NONE):
                     _GEMM_CHOICE[key] = choice
                 if choice == "gemm8":
                     out = _hip_ops.gemm8_bf16(x2, weight, bias, act)
+                elif choice == "gemm16":
+                    out = _hip_ops.gemm16_bf16(x2, weight, bias, act)
                 elif choice == "blt_fused":
                     out = _blt_fused_path(x2, weight, bias, act)
                 else:

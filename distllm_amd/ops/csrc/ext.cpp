@@ -38,6 +38,8 @@ void launch_gemm_bf16(const void*, const void*, const void*, void*, int, int,
                       int, int, hipStream_t);
 void launch_gemm8_bf16(const void*, const void*, const void*, void*, int, int,
                        int, int, hipStream_t);
+void launch_gemm16_bf16(const void*, const void*, const void*, void*, int,
+                        int, int, int, hipStream_t);
 void launch_decode_gemm(const void*, const void*, const void*, void*, void*,
                         int, int, int, int, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, const float*,
@@ -177,6 +179,21 @@ torch::Tensor gemm8_bf16(torch::Tensor A, torch::Tensor B,
   launch_gemm8_bf16(A.data_ptr(), B.data_ptr(),
                     bias ? bias->data_ptr() : nullptr, C.data_ptr(), M, N, K,
                     (int)act, cur_stream());
+  return C;
+}
+
+torch::Tensor gemm16_bf16(torch::Tensor A, torch::Tensor B,
+                          c10::optional<torch::Tensor> bias, int64_t act) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_BF16(A);
+  CHECK_DEV(B); CHECK_CONT(B); CHECK_BF16(B);
+  const int M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 128 == 0,
+              "gemm16 requires M,N % 256 == 0 and K % 128 == 0");
+  auto C = torch::empty({M, N}, A.options());
+  launch_gemm16_bf16(A.data_ptr(), B.data_ptr(),
+                     bias ? bias->data_ptr() : nullptr, C.data_ptr(), M, N,
+                     K, (int)act, cur_stream());
   return C;
 }
 
@@ -372,6 +389,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adjacent_cosine", &adjacent_cosine);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("gemm8_bf16", &gemm8_bf16);
+  m.def("gemm16_bf16", &gemm16_bf16);
   m.def("decode_gemm", &decode_gemm);
   m.def("attn_fwd", &attn_fwd);
   m.def("rope_apply", &rope_apply);

@@ -277,3 +277,43 @@ def test_blt_fused_epilogue(device, act, bias):
     out = _blt_fused_path(x, w, b, act)
     ref = torch_ref.gemm(x, w, b, act=act)
     _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("act", [0, 2, 3])  # none, gelu_erf, silu
+def test_gemm16_vs_ref(device, act):
+    """The 8-phase 256^2 flagship GEMM (gemm16.hip) vs fp32 reference,
+    flagship-like K-shallow shape, asymmetric data, fused epilogue."""
+    from distllm_amd.ops import _hip_ops
+
+    torch.manual_seed(20 + act)
+    M, N, K = 512, 768, 768
+    A = torch.randn(M, K, device=device, dtype=torch.bfloat16) * 0.5
+    B = torch.randn(N, K, device=device, dtype=torch.bfloat16) * 0.5
+    bias = torch.randn(N, device=device, dtype=torch.bfloat16)
+    C = _hip_ops.gemm16_bf16(A, B, bias, act)
+    ref = torch_ref.gemm(A, B, bias, act=act)
+    _bf16_close(C, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gemm16_deep_k_no_bias(device):
+    from distllm_amd.ops import _hip_ops
+
+    torch.manual_seed(24)
+    M, N, K = 256, 512, 3072
+    A = torch.randn(M, K, device=device, dtype=torch.bfloat16) * 0.3
+    B = torch.randn(N, K, device=device, dtype=torch.bfloat16) * 0.3
+    C = _hip_ops.gemm16_bf16(A, B, None, 0)
+    ref = torch_ref.gemm(A, B, None, 0)
+    _bf16_close(C, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gemm8_vs_ref(device):
+    from distllm_amd.ops import _hip_ops
+
+    torch.manual_seed(25)
+    M, N, K = 512, 768, 768
+    A = torch.randn(M, K, device=device, dtype=torch.bfloat16) * 0.5
+    B = torch.randn(N, K, device=device, dtype=torch.bfloat16) * 0.5
+    C = _hip_ops.gemm8_bf16(A, B, None, 0)
+    ref = torch_ref.gemm(A, B, None, 0)
+    _bf16_close(C, ref, atol=5e-2, rtol=5e-2)
