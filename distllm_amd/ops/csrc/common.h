@@ -77,15 +77,41 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* lds_scratch) {
     }                                                                         \
   } while (0)
 
+// Branch-free transcendentals on the native v_exp_f32 / v_rcp_f32
+// instructions. libm tanhf/erff/expf compile to branchy range-checked
+// slow paths (one s_and_saveexec waterfall per element — measured 37%
+// of a fused-GELU GEMM's time); these stay straight-line VALU.
+__device__ __forceinline__ float fast_exp(float x) {
+  return __builtin_amdgcn_exp2f(x * 1.4426950408889634f);
+}
+__device__ __forceinline__ float fast_rcp(float x) {
+  return __builtin_amdgcn_rcpf(x);  // ~1 ulp: plenty for bf16 outputs
+}
+__device__ __forceinline__ float fast_tanh(float u) {
+  // 1 - 2/(e^{2u}+1); exp2 under/overflow saturates to +-1 gracefully
+  const float e = __builtin_amdgcn_exp2f(u * 2.885390081777927f);
+  return 1.f - 2.f * fast_rcp(e + 1.f);
+}
+__device__ __forceinline__ float fast_erf(float z) {
+  // Abramowitz-Stegun 7.1.26 (max abs err 1.5e-7), branch-free
+  const float a = fabsf(z);
+  const float t = fast_rcp(fmaf(0.3275911f, a, 1.f));
+  float p = fmaf(1.061405429f, t, -1.453152027f);
+  p = fmaf(p, t, 1.421413741f);
+  p = fmaf(p, t, -0.284496736f);
+  p = fmaf(p, t, 0.254829592f);
+  const float y = 1.f - p * t * fast_exp(-a * a);
+  return copysignf(y, z);
+}
 __device__ __forceinline__ float fast_gelu(float x) {
-  // tanh approximation (matches BERT/HF "gelu" within bf16 tolerance)
+  // tanh approximation (matches BERT/HF "gelu_new" semantics)
   const float k0 = 0.7978845608028654f;  // sqrt(2/pi)
   const float k1 = 0.044715f;
-  return 0.5f * x * (1.f + tanhf(k0 * (x + k1 * x * x * x)));
+  return 0.5f * x * (1.f + fast_tanh(k0 * (x + k1 * x * x * x)));
 }
 __device__ __forceinline__ float gelu_erf(float x) {
-  return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
+  return 0.5f * x * (1.f + fast_erf(x * 0.70710678118654752f));
 }
 __device__ __forceinline__ float silu(float x) {
-  return x / (1.f + __expf(-x));
+  return x * fast_rcp(1.f + fast_exp(-x));
 }

@@ -40,7 +40,7 @@ sample_kernel(const float* __restrict__ logits,
   float s = 0.f;
   for (int i = threadIdx.x; i < V; i += 256) {
     float l = row[i];
-    if (l >= cut) s += __expf((l - mx) * invT);
+    if (l >= cut) s += fast_exp((l - mx) * invT);
   }
   psum[threadIdx.x] = s;
   __syncthreads();
@@ -64,7 +64,7 @@ sample_kernel(const float* __restrict__ logits,
     for (int i = threadIdx.x; i < V; i += 256) {
       float l = row[i];
       if (l >= cut) {
-        acc += __expf((l - mx) * invT);
+        acc += fast_exp((l - mx) * invT);
         if (acc > target) { out[b] = i; break; }
       }
     }

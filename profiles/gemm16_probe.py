@@ -33,7 +33,22 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--m", type=int, default=262144)
     ap.add_argument("--check", action="store_true")
+    ap.add_argument("--only", type=str, default="",
+                    help="run just gemm16 on ONE shape (qkv/ffn2/...) "
+                         "a few iters — for rocprofv3 wrapping")
     args = ap.parse_args()
+    if args.only:
+        shapes = {"qkv": (2304, 768, 0), "attn_out": (768, 768, 0),
+                  "ffn1": (3072, 768, 2), "ffn2": (768, 3072, 0)}
+        n, k, act = shapes[args.only]
+        A = torch.randn(args.m, k, device="cuda",
+                        dtype=torch.bfloat16) * 0.3
+        B = torch.randn(n, k, device="cuda", dtype=torch.bfloat16) * 0.3
+        from distllm_amd.ops import _hip_ops as ho
+        for _ in range(5):
+            ho.gemm16_bf16(A, B, None, act)
+        torch.cuda.synchronize()
+        return
     assert torch.cuda.is_available() and ops.HAVE_HIP
     M = args.m
     shapes = [
@@ -51,8 +66,10 @@ def main():
         tf = 2.0 * m * n * k / 1e12
 
         if args.check:
-            got = _hip_ops.gemm16_bf16(A[:512], B, bias, act)
-            ref = torch_ref.gemm(A[:512], B, bias, act=act)
+            # 32768 rows -> nwg > 256 so persistent multi-tile runs and
+            # the cross-tile prefetch path are exercised
+            got = _hip_ops.gemm16_bf16(A[:32768], B, bias, act)
+            ref = torch_ref.gemm(A[:32768], B, bias, act=act)
             err = (got.float() - ref.float()).abs()
             tol = 5e-2 + 5e-2 * ref.float().abs()
             bad = (err > tol).float().mean().item()
