@@ -1,4 +1,4 @@
-// Persistent 8-phase CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T
+// 8-phase deep-pipelined CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T
 // (+fused bias/activation). The gemm8.hip successor targeting the
 // K-shallow flagship encoder shapes (M=262144, N in {768,2304,3072},
 // K in {768,3072}) where hipBLASLt previously owned ~52% of the
@@ -7,36 +7,37 @@
 //
 // Structure per the guide's 256^2 8-phase template
 // (cdna_hip_programming.md §5 "The 256² 8-phase template"), own
-// implementation, made PERSISTENT for the K-shallow regime:
+// implementation and schedule:
 //   - 256x256x64 tile, 512 threads = 8 waves as 2(M) x 4(N); per-wave
 //     output 128x64 (8 mi x 4 ni fragments of mfma_f32_16x16x32_bf16).
-//   - one loop iteration consumes TWO K-tiles (8 phases); each phase:
-//     {ds_read fragment subtile; 0-2 half-tile glds prefetch;
-//      s_barrier; lgkmcnt(0); s_setprio(1); 16 MFMA; s_setprio(0);
-//      s_barrier}.
-//   - ZIGZAG quadrant order {q(mi0-3,ni01), q(mi0-3,ni23),
-//     q(mi4-7,ni23), q(mi4-7,ni01)}: at most one A-subtile (32 VGPR)
-//     plus both B-subtiles (32) live at once -> 242 VGPR, no spills,
-//     2 waves/SIMD.
-//   - counted s_waitcnt vmcnt(6) ONCE per K-tile: 3 half-tiles
-//     (2 glds each) stay in flight across every barrier (T3+T4).
-//   - PERSISTENT: grid = min(#tiles, 256); each workgroup owns a
-//     contiguous run of output tiles. During the last K-iteration the
-//     steady-state stage slots are redirected to the NEXT output
-//     tile's first K-tile (its 64 KiB first-touch is hidden under the
-//     current tile's MFMAs), killing the ~10k-cycle cold prologue
-//     that a 1-block-per-CU launch pays per tile at K=768.
-//   - LDS map: buf0 images in the first 64 KiB, buf1 in the second;
-//     the epilogue's per-wave transpose slab lives in the buf1 region
-//     so in-flight next-tile glds (always into buf0) are never
-//     drained. The slab epilogue is wave-local: s_waitcnt lgkmcnt(0)
-//     ordering only, no workgroup barriers, and each lane packs 8 f32
-//     into ONE 16-byte dwordx4 store per row pass (T21).
+//   - one iteration consumes TWO K-tiles (8 phases); ZIGZAG quadrant
+//     order {q(mi0-3,ni01), q(mi0-3,ni23), q(mi4-7,ni23),
+//     q(mi4-7,ni01)}: at most one A-subtile (32 VGPR) plus both
+//     B-subtiles (32) live at once -> no spills at 2 waves/SIMD.
+//   - ONE barrier per phase: {ds_read subtile; glds stage;
+//     s_barrier; lgkmcnt(0); s_setprio(1); 16 MFMA; s_setprio(0)}.
+//     PMC on the two-barrier variant showed 37% of wave cycles parked
+//     (SQ_WAIT_ANY); the post-MFMA barrier is dropped by DEFERRING
+//     every stage slot to >= 2 barriers after its target image's last
+//     read: any wave issuing the stage has passed BAR(p-1), and every
+//     wave's reads from 2+ phases ago completed before its own
+//     lgkmcnt(0)+MFMA which precede BAR(p-1) — placement-independent.
+//   - stage slots (tile pair t,t+1 consuming bufs 0,1):
+//       ph0: B1(t+1)+A0(t+1) -> buf1   ph1: A1(t+1) -> buf1
+//       ph3: B0(t+2) -> buf0           ph4: B1(t+2)+A0(t+2) -> buf0
+//       ph5: A1(t+2) -> buf0           ph7: B0(t+3) -> buf1
+//     counted s_waitcnt vmcnt(2) ONCE per K-tile (end of ph3 and ph7)
+//     leaves exactly the newest half-tile in flight (T3+T4).
 //   - XOR swizzle (chunk ^ (row&7)) on the glds SOURCE address and the
 //     ds_read offset (guide rule 21 / T2).
+//   - T1 bijective XCD swizzle, tm-major so the N-tiles sharing an
+//     A-panel land on one XCD's L2.
+//   - epilogue: per-wave padded-LDS transpose rounds (wave-local
+//     lgkmcnt ordering, no workgroup barriers); each lane packs 8 f32
+//     into ONE 16-byte dwordx4 store per row pass (T21), bias +
+//     branch-free GELU/SiLU fused.
 // Requires M%256==0, N%256==0, K%128==0 (host falls back otherwise).
 #include "common.h"
-#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
@@ -60,23 +61,28 @@ __global__ void __launch_bounds__(512, 2)
 gemm16_kernel(const unsigned short* __restrict__ A,
               const unsigned short* __restrict__ B,
               const unsigned short* __restrict__ bias,
-              unsigned short* __restrict__ C, int M, int N, int K,
-              int tiles_per_wg) {
+              unsigned short* __restrict__ C, int M, int N, int K) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;  // 0..7
   const int mw = wid >> 2;           // M half (0/1): rows mw*128..+127
   const int nw = wid & 3;            // N quarter: cols nw*64..+63
 
-  const int ntn = N / G16_BN;
-  const int nwg = (M / G16_BM) * ntn;
-  const int t0 = blockIdx.x * tiles_per_wg;
-  const int t_end = min(t0 + tiles_per_wg, nwg);
-  if (t0 >= nwg) return;
+  // T1 bijective XCD swizzle (tm-major tile order underneath)
+  const int ntm = M / G16_BM, ntn = N / G16_BN;
+  const int nwg = ntm * ntn;
+  int bid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, j = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + j;
+  }
+  const int tm = bid / ntn, tn = bid % ntn;
 
-  // LDS: buf-major so buf1's 64 KiB can host the epilogue slab while
-  // next-tile prefetches land in buf0.
+  const unsigned short* Atile = A + (long)tm * G16_BM * K;
+  const unsigned short* Btile = B + (long)tn * G16_BN * K;
+
   auto img = [&](int op, int buf, int half) {
-    return smem16 + ((buf * 2 + op) * 2 + half) * G16_HTILE_B;
+    return smem16 + ((op * 2 + buf) * 2 + half) * G16_HTILE_B;
   };
 
   const int r_in = lane >> 3;   // 0..7 within the 8-row piece
@@ -105,12 +111,27 @@ gemm16_kernel(const unsigned short* __restrict__ A,
         image + g16_img_off(row16 * 16 + fr_row, kk * 4 + fr_c));
   };
 
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
   const int bh = nw >> 1;         // wave's B half image
   const int br16 = (nw & 1) * 4;  // local row16 base within it
   const int nt = K / G16_BK;      // even (host guarantees K%128==0)
-  const int niters = nt / 2;
 
-  f32x4 acc[8][4];
+  // ---- prologue: tile 0 complete + B0(1); steady state stages the
+  // rest of tile 1 at ph0/ph1 of the first iteration ----------------
+  stage_half(Atile, 0, 0, 0, 0);
+  stage_half(Atile, 0, 0, 0, 1);
+  stage_half(Btile, 0, 1, 0, 0);
+  stage_half(Btile, 0, 1, 0, 1);
+  stage_half(Btile, G16_BK, 1, 1, 0);
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");  // tile 0 landed
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
   bf16x8 af[8];   // current A subtile: 4 mi x 2 kk
   bf16x8 b01[4];  // ni 0..1 x kk 0..1 (live the whole tile)
   bf16x8 b23[4];  // ni 2..3 x kk 0..1
@@ -144,204 +165,143 @@ gemm16_kernel(const unsigned short* __restrict__ A,
         dst[ni * 2 + kk] = read_frag(Bi, br16 + nih * 2 + ni, kk);
   };
 
-#define G16_BAR1()                                       \
+// one barrier per phase, placed between {reads+stage} and {MFMA}
+#define G16_BAR()                                        \
   __builtin_amdgcn_s_barrier();                          \
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
-#define G16_BAR2()                                       \
-  asm volatile("" ::: "memory");                         \
-  __builtin_amdgcn_s_barrier();                          \
-  asm volatile("" ::: "memory")
 
   const char* A0c = (const char*)img(0, 0, mw);  // wave's A half, buf0
   const char* B0c = (const char*)img(1, 0, bh);
   const char* A1c = (const char*)img(0, 1, mw);  // buf1
   const char* B1c = (const char*)img(1, 1, bh);
 
-  auto tile_ptrs = [&](int otile, const unsigned short*& At,
-                       const unsigned short*& Bt) {
-    const int tm = otile / ntn, tn = otile % ntn;
-    At = A + (long)tm * G16_BM * K;
-    Bt = B + (long)tn * G16_BN * K;
-  };
+  const int niters = nt / 2;
+  for (int it = 0; it < niters; ++it) {
+    const int t = 2 * it;  // first K-tile of the pair (buf 0)
+    const long kT1 = (long)(t + 1) * G16_BK;
+    const long kT2 = (long)(t + 2) * G16_BK;
+    const long kT3 = (long)(t + 3) * G16_BK;
+    const bool hT2 = t + 2 < nt, hT3 = t + 3 < nt;
 
-  const unsigned short *Atile, *Btile, *Anext = nullptr, *Bnext = nullptr;
-  tile_ptrs(t0, Atile, Btile);
+    // ======== K-tile t (buf 0); zigzag q00->q01->q11->q10 ========
+    // ph0: read A(mi0-3)+B(ni0-1); stage B1(t+1)+A0(t+1) -> buf1
+    rd_a(A0c, 0);
+    rd_b(B0c, 0, b01);
+    stage_half(Btile, kT1, 1, 1, 1);
+    stage_half(Atile, kT1, 0, 1, 0);
+    G16_BAR();
+    mfma_quad(0, 0, af, b01);
 
-  // cold prologue for the FIRST tile only: tile0 -> buf0
-  stage_half(Atile, 0, 0, 0, 0);
-  stage_half(Atile, 0, 0, 0, 1);
-  stage_half(Btile, 0, 1, 0, 0);
-  stage_half(Btile, 0, 1, 0, 1);
+    // ph1: read B(ni2-3); stage A1(t+1) -> buf1
+    rd_b(B0c, 1, b23);
+    stage_half(Atile, kT1, 0, 1, 1);
+    G16_BAR();
+    mfma_quad(0, 2, af, b23);
 
-  for (int ot = t0; ot < t_end; ++ot) {
-    const bool last = ot + 1 >= t_end;
-    if (!last) tile_ptrs(ot + 1, Anext, Bnext);
-    // entry stage: K-tile 1 partial {B0,B1,A0} -> buf1 (tile 0 is in
-    // buf0 already: cold prologue above or the previous tile's tail)
-    stage_half(Btile, G16_BK, 1, 1, 0);
-    stage_half(Btile, G16_BK, 1, 1, 1);
-    stage_half(Atile, G16_BK, 0, 1, 0);
-    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile 0 landed
+    // ph2: read A(mi4-7)
+    rd_a(A0c, 1);
+    G16_BAR();
+    mfma_quad(4, 2, af, b23);
+
+    // ph3: stage B0(t+2) -> buf0 (B(buf0) last read ph1: 2 bars ago);
+    //      counted wait ensures tile t+1 fully landed for ph4
+    if (hT2) stage_half(Btile, kT2, 1, 0, 0);
     __builtin_amdgcn_s_barrier();
+    mfma_quad(4, 0, af, b01);
+    if (hT2)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // tile boundary (keeps ph4 reads
+    asm volatile("" ::: "memory"); // behind every wave's ph3 MFMA srcs)
 
-#pragma unroll
-    for (int i = 0; i < 8; ++i)
-#pragma unroll
-      for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-    for (int it = 0; it < niters; ++it) {
-      const int t = 2 * it;  // first K-tile of the pair (buf 0)
-      const long kT1 = (long)(t + 1) * G16_BK;
-      const bool hT2 = t + 2 < nt, hT3 = t + 3 < nt;
-      // stage sources for the t+2 / t+3 slots: fall over to the next
-      // output tile's K-tiles 0/1 when past the end of K
-      const unsigned short* A2 = hT2 ? Atile : Anext;
-      const unsigned short* B2 = hT2 ? Btile : Bnext;
-      const long kT2 = hT2 ? (long)(t + 2) * G16_BK : 0;
-      const long kT3 = (long)(t + 3) * G16_BK;
-      // t+2 slots target buf0 in steady state; when they carry the
-      // next tile's K-tile 0 they ALSO target buf0 (same images).
-      // t+3 slots target buf1; the next tile's K-tile-1 halves are
-      // staged at the next tile's entry instead (buf1 hosts the
-      // epilogue slab in between), so tail staging never uses the
-      // t+3 slots.
-      const bool s2 = hT2 || !last;   // skip tail stages on final tile
-      const bool s3t = hT3;
-
-      // ======== K-tile t (buf 0); zigzag q00->q01->q11->q10 ========
-      // ph0: read A(mi0-3)+B(ni0-1); stage A1(t+1) -> buf1
-      rd_a(A0c, 0);
-      rd_b(B0c, 0, b01);
-      stage_half(Atile, kT1, 0, 1, 1);
-      G16_BAR1();
-      mfma_quad(0, 0, af, b01);
-      G16_BAR2();
-
-      // ph1: read B(ni2-3)
-      rd_b(B0c, 1, b23);
-      G16_BAR1();
-      mfma_quad(0, 2, af, b23);
-      G16_BAR2();
-
-      // ph2: read A(mi4-7); stage B0(t+2) -> buf0
-      rd_a(A0c, 1);
-      if (s2) stage_half(B2, kT2, 1, 0, 0);
-      G16_BAR1();
-      mfma_quad(4, 2, af, b23);
-      G16_BAR2();
-
-      // ph3: stage B1(t+2), A0(t+2) -> buf0; counted wait
-      if (s2) {
-        stage_half(B2, kT2, 1, 0, 1);
-        stage_half(A2, kT2, 0, 0, 0);
-      }
-      __builtin_amdgcn_s_barrier();
-      mfma_quad(4, 0, af, b01);
-      if (s2)
-        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-      else
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      G16_BAR2();
-
-      // ======== K-tile t+1 (buf 1) ========
-      // ph4: read A(mi0-3)+B(ni0-1); stage A1(t+2) -> buf0
-      rd_a(A1c, 0);
-      rd_b(B1c, 0, b01);
-      if (s2) stage_half(A2, kT2, 0, 0, 1);
-      G16_BAR1();
-      mfma_quad(0, 0, af, b01);
-      G16_BAR2();
-
-      // ph5: read B(ni2-3)
-      rd_b(B1c, 1, b23);
-      G16_BAR1();
-      mfma_quad(0, 2, af, b23);
-      G16_BAR2();
-
-      // ph6: read A(mi4-7); stage B0(t+3) -> buf1
-      rd_a(A1c, 1);
-      if (s3t) stage_half(Btile, kT3, 1, 1, 0);
-      G16_BAR1();
-      mfma_quad(4, 2, af, b23);
-      G16_BAR2();
-
-      // ph7: stage B1(t+3), A0(t+3) -> buf1; counted wait
-      if (s3t) {
-        stage_half(Btile, kT3, 1, 1, 1);
-        stage_half(Atile, kT3, 0, 1, 0);
-      }
-      __builtin_amdgcn_s_barrier();
-      mfma_quad(4, 0, af, b01);
-      if (s3t)
-        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-      else if (!s2)  // truly last K-iteration of the last tile
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      // else: next-tile K-tile-0 halves stay in flight (buf0 only)
-      G16_BAR2();
+    // ======== K-tile t+1 (buf 1) ========
+    // ph4: read A(mi0-3)+B(ni0-1); stage B1(t+2)+A0(t+2) -> buf0
+    rd_a(A1c, 0);
+    rd_b(B1c, 0, b01);
+    if (hT2) {
+      stage_half(Btile, kT2, 1, 0, 1);
+      stage_half(Atile, kT2, 0, 0, 0);
     }
+    G16_BAR();
+    mfma_quad(0, 0, af, b01);
 
-    // ---- epilogue: wave-local transpose slab in the buf1 region ----
-    // (in-flight next-tile glds target buf0; nothing here drains them:
-    // no __syncthreads — its fence would emit vmcnt(0) while LDS-DMA
-    // is pending — only per-wave lgkmcnt ordering.)
-    {
-      constexpr int SLAB_W = 68;  // 64 + 4 f32 pad
-      float* slab = reinterpret_cast<float*>(smem16 + 4 * G16_HTILE_B) +
-                    wid * 16 * SLAB_W;
-      const int tm = ot / ntn, tn = ot % ntn;
-      const int grow0 = tm * G16_BM + mw * 128;
-      const int gcol0 = tn * G16_BN + nw * 64;
-      const int c8 = lane & 7;
-      const int pr = lane >> 3;
+    // ph5: read B(ni2-3); stage A1(t+2) -> buf0
+    rd_b(B1c, 1, b23);
+    if (hT2) stage_half(Atile, kT2, 0, 0, 1);
+    G16_BAR();
+    mfma_quad(0, 2, af, b23);
 
-      float bv[8];
-      if (HAS_BIAS) {
-        const s16x8 raw =
-            *reinterpret_cast<const s16x8*>(bias + gcol0 + c8 * 8);
-        bf8_to_f32(raw, bv);
-      }
+    // ph6: read A(mi4-7)
+    rd_a(A1c, 1);
+    G16_BAR();
+    mfma_quad(4, 2, af, b23);
 
-#pragma unroll
-      for (int rd = 0; rd < 8; ++rd) {  // one 16-row mi fragment each
-#pragma unroll
-        for (int ni = 0; ni < 4; ++ni) {
-          const f32x4 v = acc[rd][ni];
-          const int col = ni * 16 + (lane & 15);
-          const int row0 = (lane >> 4) * 4;
-#pragma unroll
-          for (int r = 0; r < 4; ++r)
-            slab[(row0 + r) * SLAB_W + col] = v[r];
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-#pragma unroll
-        for (int p = 0; p < 2; ++p) {
-          const int r = p * 8 + pr;
-          unsigned short o[8];
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            float x = slab[r * SLAB_W + c8 * 8 + e];
-            if (HAS_BIAS) x += bv[e];
-            if (ACT == G16_GELU_TANH) x = fast_gelu(x);
-            else if (ACT == G16_GELU_ERF) x = gelu_erf(x);
-            else if (ACT == G16_SILU) x = silu(x);
-            o[e] = f2bf(x);
-          }
-          *reinterpret_cast<i32x4*>(
-              C + (long)(grow0 + rd * 16 + r) * N + gcol0 + c8 * 8) =
-              *reinterpret_cast<const i32x4*>(o);
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      }
-    }
-    // all waves done with the buf1-region slab before the next tile's
-    // entry stages overwrite buf1 images
+    // ph7: stage B0(t+3) -> buf1; counted wait for tile t+2
+    if (hT3) stage_half(Btile, kT3, 1, 1, 0);
+    __builtin_amdgcn_s_barrier();
+    mfma_quad(4, 0, af, b01);
+    if (hT3)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else if (hT2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     asm volatile("" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();  // staging quiet before LDS becomes the slab
 
-    Atile = Anext;
-    Btile = Bnext;
+  // ---- epilogue: per-wave transpose slab, one 16-row round per mi
+  // fragment; wave-local lgkmcnt ordering only; each lane emits one
+  // dwordx4 store per row pass (T21) -------------------------------
+  {
+    constexpr int SLAB_W = 68;  // 64 + 4 f32 pad
+    float* slab = reinterpret_cast<float*>(smem16) + wid * 16 * SLAB_W;
+    const int grow0 = tm * G16_BM + mw * 128;
+    const int gcol0 = tn * G16_BN + nw * 64;
+    const int c8 = lane & 7;
+    const int pr = lane >> 3;
+
+    float bv[8];
+    if (HAS_BIAS) {
+      const s16x8 raw =
+          *reinterpret_cast<const s16x8*>(bias + gcol0 + c8 * 8);
+      bf8_to_f32(raw, bv);
+    }
+
+#pragma unroll
+    for (int rd = 0; rd < 8; ++rd) {  // one 16-row mi fragment each
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const f32x4 v = acc[rd][ni];
+        const int col = ni * 16 + (lane & 15);
+        const int row0 = (lane >> 4) * 4;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          slab[(row0 + r) * SLAB_W + col] = v[r];
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int p = 0; p < 2; ++p) {
+        const int r = p * 8 + pr;
+        unsigned short o[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float x = slab[r * SLAB_W + c8 * 8 + e];
+          if (HAS_BIAS) x += bv[e];
+          if (ACT == G16_GELU_TANH) x = fast_gelu(x);
+          else if (ACT == G16_GELU_ERF) x = gelu_erf(x);
+          else if (ACT == G16_SILU) x = silu(x);
+          o[e] = f2bf(x);
+        }
+        *reinterpret_cast<i32x4*>(
+            C + (long)(grow0 + rd * 16 + r) * N + gcol0 + c8 * 8) =
+            *reinterpret_cast<const i32x4*>(o);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
   }
 }
 
@@ -349,20 +309,11 @@ extern "C" void launch_gemm16_bf16(const void* A, const void* B,
                                    const void* bias, void* C, int M, int N,
                                    int K, int act, hipStream_t stream) {
   const int nwg = (M / G16_BM) * (N / G16_BN);
-  // DISTLLM_G16_FLAT=1: one tile per workgroup (A/B probe of the
-  // persistent scheduling; same kernel, tail staging disabled by
-  // last=true on every tile)
-  static const bool flat = [] {
-    const char* e = getenv("DISTLLM_G16_FLAT");
-    return e && e[0] == '1';
-  }();
-  const int grid = flat ? nwg : (nwg < 256 ? nwg : 256);
-  const int tpw = (nwg + grid - 1) / grid;
 #define G16D(ACT, HB)                                                         \
-  hipLaunchKernelGGL((gemm16_kernel<ACT, HB>), dim3(grid), dim3(512),         \
+  hipLaunchKernelGGL((gemm16_kernel<ACT, HB>), dim3(nwg), dim3(512),          \
                      G16_LDS, stream, (const unsigned short*)A,               \
                      (const unsigned short*)B, (const unsigned short*)bias,   \
-                     (unsigned short*)C, M, N, K, tpw)
+                     (unsigned short*)C, M, N, K)
   const bool hb = bias != nullptr;
   switch (act) {
     case G16_GELU_TANH: if (hb) G16D(1, true); else G16D(1, false); break;
