@@ -103,6 +103,11 @@ class Esm2Encoder(nn.Module, _RotaryMixin):
         ids = input_ids.long()
         B, S = ids.shape
         x = self.word_emb[ids]
+        import os
+        if (x.is_cuda and ops.HAVE_HIP and attention_mask is not None
+                and not os.environ.get("DISTLLM_NO_VARLEN")
+                and not bool(attention_mask.all())):
+            return self._forward_packed(ids, attention_mask, x)
         if self.token_dropout:
             masked = ids == ESM_MASK_TOKEN
             x = x.masked_fill(masked.unsqueeze(-1), 0.0)
@@ -146,6 +151,61 @@ class Esm2Encoder(nn.Module, _RotaryMixin):
             x = x + ops.linear(ff, layer.w2, layer.b2)
         return ops.layer_norm(x, self.final_ln_w, self.final_ln_b,
                               eps=d.layer_norm_eps)
+
+    @torch.no_grad()
+    def _forward_packed(self, ids, attention_mask, x):
+        """Varlen path for mixed-length protein batches: pack the valid
+        tokens ([T, H]), run every GEMM/norm on packed rows (zero padded
+        FLOPs) and attention through the cu_seqlens flash kernel
+        (reference used faesm flash-attn varlen,
+        distllm/embed/encoders/esm2.py:41-55). Returns padded [B,S,H]
+        with zeros at padding (poolers mask them out)."""
+        d = self.dims
+        B, S = ids.shape
+        maskb = attention_mask.bool()
+        lens = attention_mask.int().sum(1)
+        cu = torch.zeros(B + 1, dtype=torch.int32, device=x.device)
+        cu[1:] = lens.cumsum(0)
+        pos = (attention_mask.int().cumsum(1) - 1).clamp(min=0)[maskb]
+        bmap = ops.varlen_block_map(cu)
+        xp = x[maskb]  # [T, H]
+        if self.token_dropout:
+            masked = (ids == ESM_MASK_TOKEN)[maskb]
+            xp = xp.masked_fill(masked.unsqueeze(-1), 0.0)
+            mask_ratio_train = 0.15 * 0.8
+            # per-sequence observed-mask ratio, broadcast to tokens
+            seg = torch.repeat_interleave(
+                torch.arange(B, device=x.device), lens.long())
+            nmask = torch.zeros(B, device=x.device).index_add_(
+                0, seg, masked.float())
+            ratio = nmask / lens.float().clamp(min=1.0)
+            scale = (1 - mask_ratio_train) / (1 - ratio).clamp(min=1e-6)
+            xp = (xp.float() * scale[seg].unsqueeze(-1)).to(xp.dtype)
+        cs, sn = self._rope(x.device)
+        nh, hd = d.num_heads, d.head_dim
+        H = d.hidden_size
+        T = xp.shape[0]
+        for layer in self.layers:
+            h = ops.layer_norm(xp, layer.ln1_w, layer.ln1_b,
+                               eps=d.layer_norm_eps)
+            qkv = ops.linear(h, layer.wqkv, layer.bqkv)
+            q = qkv[..., :H].view(T, nh, hd)
+            k = qkv[..., H: 2 * H].view(T, nh, hd)
+            v = qkv[..., 2 * H:].view(T, nh, hd).contiguous()
+            q = ops.rope_apply(q.contiguous(), cs, sn, pos)
+            k = ops.rope_apply(k.contiguous(), cs, sn, pos)
+            ctx = ops.attn_varlen(q, k, v, cu, bmap,
+                                  scale=1.0 / math.sqrt(hd))
+            xp = xp + ops.linear(ctx.reshape(T, H), layer.wo, layer.bo)
+            h = ops.layer_norm(xp, layer.ln2_w, layer.ln2_b,
+                               eps=d.layer_norm_eps)
+            ff = ops.linear(h, layer.w1, layer.b1, act=ops.ACT_GELU_ERF)
+            xp = xp + ops.linear(ff, layer.w2, layer.b2)
+        xp = ops.layer_norm(xp, self.final_ln_w, self.final_ln_b,
+                            eps=d.layer_norm_eps)
+        out = torch.zeros(B, S, H, dtype=xp.dtype, device=xp.device)
+        out[maskb] = xp
+        return out
 
 
 class EsmCLayer(nn.Module):
@@ -192,6 +252,11 @@ class EsmCEncoder(nn.Module, _RotaryMixin):
         ids = input_ids.long()
         B, S = ids.shape
         x = self.word_emb[ids]
+        import os
+        if (x.is_cuda and ops.HAVE_HIP and attention_mask is not None
+                and not os.environ.get("DISTLLM_NO_VARLEN")
+                and not bool(attention_mask.all())):
+            return self._forward_packed(ids, attention_mask, x)
         mask_f = torch.where(
             attention_mask.bool(),
             torch.zeros((), device=ids.device, dtype=torch.float32),
@@ -226,3 +291,43 @@ class EsmCEncoder(nn.Module, _RotaryMixin):
             x = x + ops.linear(ops.swiglu(gate, up), layer.wdown) * rs
         return ops.layer_norm(x, self.final_ln_w, self.final_ln_b,
                               eps=d.layer_norm_eps)
+
+    @torch.no_grad()
+    def _forward_packed(self, ids, attention_mask, x):
+        """Varlen path (see Esm2Encoder._forward_packed)."""
+        d = self.dims
+        B, S = ids.shape
+        maskb = attention_mask.bool()
+        lens = attention_mask.int().sum(1)
+        cu = torch.zeros(B + 1, dtype=torch.int32, device=x.device)
+        cu[1:] = lens.cumsum(0)
+        pos = (attention_mask.int().cumsum(1) - 1).clamp(min=0)[maskb]
+        bmap = ops.varlen_block_map(cu)
+        xp = x[maskb]
+        cs, sn = self._rope(x.device)
+        nh, hd = d.num_heads, d.head_dim
+        H = d.hidden_size
+        T = xp.shape[0]
+        rs = self.residue_scale
+        for layer in self.layers:
+            h = ops.layer_norm(xp, layer.ln1_w, layer.ln1_b,
+                               eps=d.layer_norm_eps)
+            qkv = ops.linear(h, layer.wqkv)
+            q = qkv[..., :H].view(T, nh, hd)
+            k = qkv[..., H: 2 * H].view(T, nh, hd)
+            v = qkv[..., 2 * H:].view(T, nh, hd).contiguous()
+            q = ops.rope_apply(q.contiguous(), cs, sn, pos)
+            k = ops.rope_apply(k.contiguous(), cs, sn, pos)
+            ctx = ops.attn_varlen(q, k, v, cu, bmap,
+                                  scale=1.0 / math.sqrt(hd))
+            xp = xp + ops.linear(ctx.reshape(T, H), layer.wo) * rs
+            h = ops.layer_norm(xp, layer.ln2_w, layer.ln2_b,
+                               eps=d.layer_norm_eps)
+            gate = ops.linear(h, layer.wgate)
+            up = ops.linear(h, layer.wup)
+            xp = xp + ops.linear(ops.swiglu(gate, up), layer.wdown) * rs
+        xp = ops.layer_norm(xp, self.final_ln_w, self.final_ln_b,
+                            eps=d.layer_norm_eps)
+        out = torch.zeros(B, S, H, dtype=xp.dtype, device=xp.device)
+        out[maskb] = xp
+        return out

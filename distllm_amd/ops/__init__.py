@@ -319,6 +319,51 @@ def attn_fwd(Q, K, V, mask=None, causal=False, scale=None):
     return torch_ref.attn_fwd(Q, K, V, mask, causal, scale)
 
 
+def varlen_block_map(cu_seqlens: torch.Tensor, qblk: int = 128):
+    """Host-built (seq, qblock) map for attn_varlen: one entry per
+    128-row q block of each sequence. cu_seqlens int32 [B+1] (either
+    device). Returns (blk_seq, blk_qb) int32 on cu's device."""
+    cu = cu_seqlens.to("cpu", torch.int64)
+    lens = cu[1:] - cu[:-1]
+    nqb = torch.div(lens + qblk - 1, qblk, rounding_mode="floor")
+    blk_seq = torch.repeat_interleave(
+        torch.arange(len(lens), dtype=torch.int32), nqb
+    )
+    blk_qb = torch.cat(
+        [torch.arange(int(n), dtype=torch.int32) for n in nqb]
+    ) if len(lens) else torch.zeros(0, dtype=torch.int32)
+    dev = cu_seqlens.device
+    return blk_seq.to(dev), blk_qb.to(dev)
+
+
+def attn_varlen(Q, K, V, cu_seqlens, block_map=None, scale=None):
+    """Varlen (packed, cu_seqlens) bidirectional attention.
+    Q/K/V [total_tokens, H(KV), D]; returns [total_tokens, H, D].
+    The protein-path replacement for faesm's flash-attn varlen
+    (reference distllm/embed/encoders/esm2.py:41-55)."""
+    D = Q.shape[-1]
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if Q.is_cuda:
+        _require_hip()
+        if block_map is None:
+            block_map = varlen_block_map(cu_seqlens)
+        blk_seq, blk_qb = block_map
+        if D not in (64, 128):
+            Dp = 64 if D < 64 else 128
+            pad = Dp - D
+            O = _hip_ops.attn_varlen(
+                torch.nn.functional.pad(Q, (0, pad)).contiguous(),
+                torch.nn.functional.pad(K, (0, pad)).contiguous(),
+                torch.nn.functional.pad(V, (0, pad)).contiguous(),
+                cu_seqlens.int(), blk_seq, blk_qb, scale)
+            return O[..., :D].contiguous()
+        return _hip_ops.attn_varlen(Q.contiguous(), K.contiguous(),
+                                    V.contiguous(), cu_seqlens.int(),
+                                    blk_seq, blk_qb, scale)
+    return torch_ref.attn_varlen(Q, K, V, cu_seqlens, scale)
+
+
 def rope_apply(x, cs, sn, pos):
     if x.is_cuda:
         _require_hip()

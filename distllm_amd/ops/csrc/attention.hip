@@ -43,15 +43,24 @@ union frag_u {
   unsigned int d[4];
 };
 
-template <int D, bool CAUSAL, bool HAS_MASK, bool VDIRECT>
+// VARLEN: Q/K/V are PACKED [total_tokens, H(KV), D] (cu_seqlens int32
+// [B+1]); one workgroup per (sequence, q-block, head) via the
+// host-built blk_seq/blk_qb map — the protein/fasta path with zero
+// padded FLOPs (reference: the faesm flash-attn varlen role,
+// distllm/embed/encoders/esm2.py:41-55).
+template <int D, bool CAUSAL, bool HAS_MASK, bool VDIRECT,
+          bool VARLEN = false>
 __global__ void __launch_bounds__(256, 1)
 attn_fwd_kernel(const unsigned short* __restrict__ Q,
                 const unsigned short* __restrict__ Kt,
                 const unsigned short* __restrict__ Vt,
                 const float* __restrict__ maskp,
-                unsigned short* __restrict__ O, int Bsz, int SQ, int SKV,
+                unsigned short* __restrict__ O, int Bsz, int SQa, int SKVa,
                 int H, int HKV, float scale, long bstr_q, long tstr_q,
-                long bstr_k, long tstr_k) {
+                long bstr_k, long tstr_k,
+                const int* __restrict__ cu = nullptr,
+                const int* __restrict__ blk_seq = nullptr,
+                const int* __restrict__ blk_qb = nullptr) {
   constexpr int NWAVE = 4;  // waves per workgroup (8 measured slower)
   constexpr int NTHR = NWAVE * 64;
   constexpr int QBLK = NWAVE * QWAVE;       // q rows per workgroup
@@ -66,27 +75,40 @@ attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int h = lane >> 5;   // half index within wave
   const int qcol = lane & 31;
 
-  const int nqb = (SQ + QBLK - 1) / QBLK;
-  // T1 bijective XCD swizzle: keep the q-blocks sharing one (b, head)'s
-  // K/V on one XCD's L2 (measured +12% at D=128/GQA, neutral at D=64).
-  int bid = blockIdx.x;
-  {
-    const int nwg = Bsz * H * nqb;
-    const int q = nwg / 8, r = nwg % 8;
-    const int xcd = bid % 8, j = bid / 8;
-    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + j;
+  int qb, b, head;
+  if (VARLEN) {
+    const int i = blockIdx.x / H;
+    head = blockIdx.x % H;
+    b = blk_seq[i];
+    qb = blk_qb[i];
+  } else {
+    const int nqb0 = (SQa + QBLK - 1) / QBLK;
+    // T1 bijective XCD swizzle: keep the q-blocks sharing one
+    // (b, head)'s K/V on one XCD's L2 (+12% at D=128/GQA).
+    int bid = blockIdx.x;
+    {
+      const int nwg = Bsz * H * nqb0;
+      const int q = nwg / 8, r = nwg % 8;
+      const int xcd = bid % 8, j = bid / 8;
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + j;
+    }
+    qb = bid % nqb0;
+    const int bh = bid / nqb0;
+    head = bh % H;
+    b = bh / H;
   }
-  const int qb = bid % nqb;
-  const int bh = bid / nqb;
-  const int head = bh % H;
-  const int b = bh / H;
+  const int SQ = VARLEN ? cu[b + 1] - cu[b] : SQa;
+  const int SKV = VARLEN ? SQ : SKVa;
   const int kvhead = head / (H / HKV);
 
   const long ostride = (long)H * D;
-  const unsigned short* Qb = Q + (long)b * bstr_q + (long)head * D;
-  const unsigned short* Kb = Kt + (long)b * bstr_k + (long)kvhead * D;
-  const unsigned short* Vb = Vt + (long)b * bstr_k + (long)kvhead * D;
-  unsigned short* Ob = O + ((long)b * SQ) * ostride + (long)head * D;
+  const long qbase = VARLEN ? (long)cu[b] * tstr_q : (long)b * bstr_q;
+  const long kbase = VARLEN ? (long)cu[b] * tstr_k : (long)b * bstr_k;
+  const unsigned short* Qb = Q + qbase + (long)head * D;
+  const unsigned short* Kb = Kt + kbase + (long)kvhead * D;
+  const unsigned short* Vb = Vt + kbase + (long)kvhead * D;
+  unsigned short* Ob = O + (VARLEN ? (long)cu[b] : (long)b * SQ) * ostride +
+                       (long)head * D;
 
   const int q0w = qb * QBLK + wid * QWAVE;
   const int qrow_abs = q0w + qcol;
@@ -424,4 +446,35 @@ extern "C" void launch_attn_fwd(const void* Q, const void* K, const void* V,
     else ATTN_DISPATCH(128, false, false);
   }
 #undef ATTN_DISPATCH
+}
+
+extern "C" void launch_attn_varlen(const void* Q, const void* K,
+                                   const void* V, void* O, const int* cu,
+                                   const int* blk_seq, const int* blk_qb,
+                                   int nblocks, int B, int H, int HKV,
+                                   int D, float scale,
+                                   hipStream_t stream) {
+  const int nwave = 4;
+  const int grid = nblocks * H;
+  const int ktile = KVBLK * D * 2, vtile = D * VROW * 2;
+  const int stage_b = 2 * ktile + 2 * vtile + 2 * KVBLK * 4;
+  const int epi_b = nwave * 32 * (D + 1) * 4;
+  const int lds = stage_b > epi_b ? stage_b : epi_b;
+  const long tq = (long)H * D, tk = (long)HKV * D;
+  static int vdirect = -1;
+  if (vdirect < 0) {
+    const char* e = getenv("DISTLLM_ATTN_VDIRECT");
+    vdirect = e ? (e[0] == '1' ? 1 : 0) : -1;
+  }
+  const int vd = (vdirect >= 0) ? vdirect : (D == 128 ? 1 : 0);
+#define AVL(DD, VD)                                                           \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, false, false, VD, true>),           \
+                     dim3(grid), dim3(nwave * 64), lds, stream,               \
+                     (const unsigned short*)Q, (const unsigned short*)K,      \
+                     (const unsigned short*)V, (const float*)nullptr,         \
+                     (unsigned short*)O, B, 0, 0, H, HKV, scale, 0L, tq,      \
+                     0L, tk, cu, blk_seq, blk_qb)
+  if (D == 64) { if (vd) AVL(64, true); else AVL(64, false); }
+  else if (D == 128) { if (vd) AVL(128, true); else AVL(128, false); }
+#undef AVL
 }

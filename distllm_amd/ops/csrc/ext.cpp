@@ -45,6 +45,9 @@ void launch_decode_gemm(const void*, const void*, const void*, void*, void*,
 void launch_attn_fwd(const void*, const void*, const void*, const float*,
                      void*, int, int, int, int, int, int, bool, float, long,
                      long, long, long, hipStream_t);
+void launch_attn_varlen(const void*, const void*, const void*, void*,
+                        const int*, const int*, const int*, int, int, int,
+                        int, int, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, const int*,
                  long, int, int, hipStream_t);
 void launch_topk_chunk(const float*, const int*, float*, int*, int, long, int,
@@ -256,6 +259,32 @@ torch::Tensor attn_fwd(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
   return O;
 }
 
+torch::Tensor attn_varlen(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
+                          torch::Tensor cu, torch::Tensor blk_seq,
+                          torch::Tensor blk_qb, double scale) {
+  // packed [T, H(KV), D] bf16; cu int32 [B+1]; blk_* int32 [nblocks]
+  CHECK_DEV(Q); CHECK_CONT(Q); CHECK_BF16(Q);
+  CHECK_DEV(K); CHECK_CONT(K); CHECK_BF16(K);
+  CHECK_DEV(V); CHECK_CONT(V); CHECK_BF16(V);
+  CHECK_DEV(cu); CHECK_CONT(cu);
+  CHECK_DEV(blk_seq); CHECK_CONT(blk_seq);
+  CHECK_DEV(blk_qb); CHECK_CONT(blk_qb);
+  TORCH_CHECK(cu.scalar_type() == torch::kInt32 &&
+              blk_seq.scalar_type() == torch::kInt32 &&
+              blk_qb.scalar_type() == torch::kInt32,
+              "cu/blk maps must be int32");
+  const int H = Q.size(1), D = Q.size(2), HKV = K.size(1);
+  const int B = cu.size(0) - 1, nblocks = blk_seq.size(0);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  TORCH_CHECK(H % HKV == 0, "GQA requires H % HKV == 0");
+  auto O = torch::empty_like(Q);
+  launch_attn_varlen(Q.data_ptr(), K.data_ptr(), V.data_ptr(), O.data_ptr(),
+                     cu.data_ptr<int>(), blk_seq.data_ptr<int>(),
+                     blk_qb.data_ptr<int>(), nblocks, B, H, HKV, D,
+                     (float)scale, cur_stream());
+  return O;
+}
+
 // ---- rope ------------------------------------------------------------------
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
                          torch::Tensor pos) {
@@ -392,6 +421,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm16_bf16", &gemm16_bf16);
   m.def("decode_gemm", &decode_gemm);
   m.def("attn_fwd", &attn_fwd);
+  m.def("attn_varlen", &attn_varlen);
   m.def("rope_apply", &rope_apply);
   m.def("topk_chunked", &topk_chunked);
   m.def("hamming_score", &hamming_score);

@@ -166,6 +166,21 @@ def attn_fwd(
     return o.permute(0, 2, 1, 3).contiguous().to(Q.dtype)
 
 
+def attn_varlen(Q, K, V, cu_seqlens, scale: float):
+    """Packed [T, H(KV), D] varlen bidirectional attention reference:
+    per-sequence slices through attn_fwd (fp32)."""
+    cu = cu_seqlens.to("cpu", torch.int64).tolist()
+    out = torch.empty_like(Q)
+    for b in range(len(cu) - 1):
+        lo, hi = cu[b], cu[b + 1]
+        if hi <= lo:
+            continue
+        o = attn_fwd(Q[lo:hi][None], K[lo:hi][None], V[lo:hi][None],
+                     None, False, scale)
+        out[lo:hi] = o[0]
+    return out
+
+
 def rope_tables(max_pos: int, D: int, theta: float = 10000.0, device="cpu"):
     inv = 1.0 / (
         theta ** (torch.arange(0, D, 2, dtype=torch.float32, device=device) / D)

@@ -217,3 +217,36 @@ def test_ref_last_token_pool_vs_naive():
     out_l = torch_ref.last_token_pool(h[:1], mask_l, left_padded=True,
                                       normalize=False)
     assert torch.allclose(out_l[0], h[0, 5])
+
+
+def test_attn_varlen_ref_matches_padded():
+    """torch_ref.attn_varlen == padded attn_fwd with key-pad mask on
+    the valid positions."""
+    import math
+
+    from distllm_amd.ops import torch_ref, varlen_block_map
+
+    torch.manual_seed(40)
+    lens = [7, 20, 1, 13]
+    B, S, H, D = len(lens), max(lens), 3, 16
+    q = torch.randn(B, S, H, D, dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, D, dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, D, dtype=torch.bfloat16)
+    maskb = torch.zeros(B, S, dtype=torch.bool)
+    for r, L in enumerate(lens):
+        maskb[r, :L] = True
+    mask_f = torch.where(maskb, 0.0, float("-inf")).float()
+    scale = 1.0 / math.sqrt(D)
+    padded = torch_ref.attn_fwd(q, k, v, mask_f, False, scale)
+
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32)
+    packed = torch_ref.attn_varlen(q[maskb], k[maskb], v[maskb], cu, scale)
+    assert torch.allclose(packed.float(), padded[maskb].float(),
+                          atol=2e-2, rtol=2e-2)
+
+    bs, bq = varlen_block_map(cu)
+    assert bs.tolist() == [0, 1, 2, 3] and bq.tolist() == [0, 0, 0, 0]
+    bs2, bq2 = varlen_block_map(
+        torch.tensor([0, 130, 386], dtype=torch.int32))
+    assert bs2.tolist() == [0, 0, 1, 1] and bq2.tolist() == [0, 1, 0, 1]
