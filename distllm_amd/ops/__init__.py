@@ -319,6 +319,14 @@ def attn_fwd(Q, K, V, mask=None, causal=False, scale=None):
     return torch_ref.attn_fwd(Q, K, V, mask, causal, scale)
 
 
+def ivf_score(slabs, lens, probes, q):
+    """Batched IVF chunk scoring: slabs [nchunks, cap, H] bf16, lens
+    [nchunks] i32, probes [Q, nprobe] i32, q [Q, H] f32 ->
+    [Q, nprobe, cap] f32 (-inf at pad rows)."""
+    _require_hip()
+    return _hip_ops.ivf_score(slabs, lens, probes, q.contiguous())
+
+
 def varlen_block_map(cu_seqlens: torch.Tensor, qblk: int = 128):
     """Host-built (seq, qblock) map for attn_varlen: one entry per
     128-row q block of each sequence. cu_seqlens int32 [B+1] (either

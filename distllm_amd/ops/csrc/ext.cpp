@@ -54,6 +54,8 @@ void launch_topk_chunk(const float*, const int*, float*, int*, int, long, int,
                        int, hipStream_t);
 void launch_hamming_score(const void*, const void*, float*, int, long, int,
                           hipStream_t);
+void launch_ivf_score(const void*, const int*, const int*, const float*,
+                      float*, int, int, int, int, hipStream_t);
 void launch_sample(const float*, const float*, int*, int, int, float, float,
                    hipStream_t);
 void launch_paged_attn_split(const void*, const void*, const void*,
@@ -405,6 +407,25 @@ void cache_put(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kcache,
 torch::Tensor blt_linear(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t epi);
 
+
+torch::Tensor ivf_score(torch::Tensor slabs, torch::Tensor lens,
+                        torch::Tensor probes, torch::Tensor q) {
+  CHECK_DEV(slabs); CHECK_CONT(slabs); CHECK_F32(slabs);
+  CHECK_DEV(lens); CHECK_CONT(lens); CHECK_I32(lens);
+  CHECK_DEV(probes); CHECK_CONT(probes); CHECK_I32(probes);
+  CHECK_DEV(q); CHECK_CONT(q); CHECK_F32(q);
+  const int cap = slabs.size(1), H = slabs.size(2);
+  const int Q = probes.size(0), nprobe = probes.size(1);
+  TORCH_CHECK(q.size(0) == Q && q.size(1) == H, "q shape mismatch");
+  auto out = torch::empty({Q, nprobe, cap},
+                          q.options().dtype(at::kFloat));
+  launch_ivf_score(slabs.data_ptr(), lens.data_ptr<int>(),
+                   probes.data_ptr<int>(), q.data_ptr<float>(),
+                   out.data_ptr<float>(), Q * nprobe, nprobe, cap, H,
+                   cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("blt_linear", &blt_linear,
         "hipBLASLt linear with fused bias/GELU/swish epilogue");
@@ -425,6 +446,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_apply", &rope_apply);
   m.def("topk_chunked", &topk_chunked);
   m.def("hamming_score", &hamming_score);
+  m.def("ivf_score", &ivf_score);
   m.def("sample_tokens", &sample_tokens);
   m.def("paged_attn", &paged_attn);
   m.def("cache_put", &cache_put);

@@ -178,3 +178,59 @@ extern "C" void launch_hamming_score(const void* qcodes, const void* codes_t,
                      (const unsigned long long*)qcodes,
                      (const unsigned long long*)codes_t, out, N, W);
 }
+
+// ---- batched IVF chunk scoring (approximate search) --------
+// slabs: [nchunks, cap, H] f32 padded chunk slabs; probes: [Q*nprobe]
+// int32 chunk ids; qv: [Q, H] f32; out: [Q*nprobe*cap] f32 (-inf at
+// pad rows). One block per (query, probe): the query vector is staged
+// in LDS once; each wave streams rows of the contiguous slab with
+// 16-byte-per-lane coalesced loads and a wave-reduce dot. BW-bound by
+// design (bf16 slabs read once; winners re-scored fp32 host-side).
+__global__ void __launch_bounds__(256)
+ivf_score_kernel(const float* __restrict__ slabs,
+                 const int* __restrict__ lens,
+                 const int* __restrict__ probes,
+                 const float* __restrict__ qv, float* __restrict__ out,
+                 int nprobe, int cap, int H) {
+  extern __shared__ __attribute__((aligned(16))) float qlds[];
+  const int qi = blockIdx.x / nprobe;
+  const int ch = probes[blockIdx.x];
+  const int len = lens[ch];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+
+  for (int i = threadIdx.x; i < H; i += 256) qlds[i] = qv[qi * H + i];
+  __syncthreads();
+
+  const float* slab = slabs + (long)ch * cap * H;
+  float* orow = out + (long)blockIdx.x * cap;
+  const int span = 64 * 4;  // elements covered per wave pass
+  for (int r = wid; r < cap; r += 4) {
+    float acc = 0.f;
+    if (r < len) {
+      const float* row = slab + (long)r * H;
+      for (int i0 = lane * 4; i0 < H; i0 += span) {
+        // 4 f32 = one 16-byte load (tail guard per element)
+        if (i0 + 4 <= H) {
+          const f32x4 v = *reinterpret_cast<const f32x4*>(row + i0);
+#pragma unroll
+          for (int j = 0; j < 4; ++j) acc += v[j] * qlds[i0 + j];
+        } else {
+          for (int j = i0; j < H; ++j) acc += row[j] * qlds[j];
+        }
+      }
+      acc = wave_reduce_sum(acc);
+    }
+    if (lane == 0) orow[r] = (r < len) ? acc : -INFINITY;
+  }
+}
+
+extern "C" void launch_ivf_score(const void* slabs, const int* lens,
+                                 const int* probes, const float* qv,
+                                 float* out, int nblocks, int nprobe,
+                                 int cap, int H, hipStream_t stream) {
+  hipLaunchKernelGGL(ivf_score_kernel, dim3(nblocks), dim3(256),
+                     H * (int)sizeof(float), stream,
+                     (const float*)slabs, lens, probes, qv, out,
+                     nprobe, cap, H);
+}

@@ -175,73 +175,111 @@ class HipFlatIndex:
         if self.search_algorithm == "ivf":
             self._build_ivf()
 
+    def _assign(self, cent: torch.Tensor) -> torch.Tensor:
+        """argmax cluster per corpus row, GEMM-chunked so the score
+        matrix never exceeds ~1 GB at 10M+ rows."""
+        rows = max(1, (1 << 28) // max(cent.shape[0], 1))
+        outs = []
+        for r0 in range(0, self.n, rows):
+            outs.append(
+                (self.corpus[r0:r0 + rows] @ cent.t()).argmax(1))
+        return torch.cat(outs)
+
     def _build_ivf(self) -> None:
-        """Simple GPU k-means (inner-product on normalized centroids);
-        corpus reordered cluster-major for contiguous candidate gathers.
-        Replaces the reference's faiss HNSW approximate option
-        (rag/search.py:237,245) with an IVF-flat design that maps to
-        GEMM + gather on this hardware."""
+        """Batched IVF: GPU k-means, then clusters split into chunks of
+        at most `cap` rows and stored as PADDED bf16 slabs
+        [nchunks, cap, dim] with per-chunk row->corpus-id maps. Search
+        scores every probed chunk with one regular kernel launch (no
+        per-query host loop) and top-ks over the padded candidate
+        matrix. Replaces the reference's faiss HNSW approximate option
+        (rag/search.py:237,245) with a design sized for 10M+ corpora
+        resident in 288 GB HBM (VERDICT.md #7)."""
         C = self.nlist or max(int(self.n ** 0.5), 1)
         C = min(C, self.n)
+        dev = self.corpus.device
         g = torch.Generator().manual_seed(0)
         init = torch.randperm(self.n, generator=g)[:C]
-        cent = self.corpus[init.to(self.corpus.device)].clone()
+        cent = self.corpus[init.to(dev)].clone()
+        ar = torch.arange(C, device=dev)
         for _ in range(10):
-            assign = (self.corpus @ cent.t()).argmax(1)
-            for c in range(C):
-                m = assign == c
-                if m.any():
-                    cent[c] = self.corpus[m].mean(0)
-        assign = (self.corpus @ cent.t()).argmax(1)
-        order = torch.argsort(assign, stable=True)
-        counts = torch.bincount(assign, minlength=C)
+            assign = self._assign(cent)
+            sums = torch.zeros_like(cent).index_add_(0, assign, self.corpus)
+            cnt = torch.bincount(assign, minlength=C).float()
+            keep = cnt > 0
+            cent[keep] = sums[keep] / cnt[keep].unsqueeze(1)
+        assign = self._assign(cent)
+        order = torch.argsort(assign.cpu(), stable=True)
+        counts = torch.bincount(assign.cpu(), minlength=C)
         offs = torch.zeros(C + 1, dtype=torch.long)
-        offs[1:] = counts.cumsum(0).cpu()
-        self.ivf_centroids = cent
-        self.ivf_order = order            # corpus row ids, cluster-major
-        self.ivf_offsets = offs           # [C+1] boundaries into order
+        offs[1:] = counts.cumsum(0)
+
+        # split clusters into <=cap-row chunks; pad each to cap
+        cap = max(64, (self.n + C - 1) // C)
+        chunk_src: list[tuple[int, int]] = []  # (start, len) into order
+        for c in range(C):
+            o0, o1 = int(offs[c]), int(offs[c + 1])
+            for s in range(o0, o1, cap):
+                chunk_src.append((s, min(cap, o1 - s)))
+        nch = len(chunk_src)
+        idx_pad = torch.full((nch, cap), -1, dtype=torch.int32)
+        for i, (s, ln) in enumerate(chunk_src):
+            idx_pad[i, :ln] = order[s:s + ln].int()
+        lens = torch.tensor([ln for _, ln in chunk_src],
+                            dtype=torch.int32)
+        # padded fp32 slabs: exact scores (bf16 slabs measured to
+        # scramble near-tie ordering on tight clusters); 2x traffic is
+        # still far below the exact path at small query counts
+        slabs = torch.zeros(nch, cap, self.dim, dtype=torch.float32,
+                            device=dev)
+        gather = idx_pad.clamp(min=0).long().to(dev)
+        slabs.copy_(
+            self.corpus[gather.reshape(-1)].reshape(nch, cap, self.dim))
+        # zero the pad rows (the clamp pointed them at corpus[0]): they
+        # must not contribute to chunk means nor produce candidates
+        valid = (torch.arange(cap, device=dev)[None, :]
+                 < lens.to(dev)[:, None])
+        slabs.mul_(valid.unsqueeze(-1))
+        # per-chunk centroid = chunk mean (valid rows only)
+        csum = slabs.float().sum(1)
+        ccent = csum / lens.to(dev).float().clamp(min=1).unsqueeze(1)
+
         self.ivf_nlist = C
-        # cluster-major slab copy: probing then reads CONTIGUOUS rows
-        # (a [len, D] @ [D] GEMV per probed cluster) instead of a random
-        # row gather — 384x less traffic than gathering candidate rows.
-        # Costs one corpus duplicate; by design for 288 GB HBM.
-        self.corpus_sorted = self.corpus[order.to(self.corpus.device)]\
-            .contiguous()
+        self.ivf_cap = cap
+        self.ivf_nchunks = nch
+        self.ivf_centroids = ccent        # [nchunks, dim] chunk means
+        self.ivf_slabs = slabs            # [nchunks, cap, dim] bf16
+        self.ivf_idx = idx_pad.to(dev)    # [nchunks, cap] int32, -1 pad
+        self.ivf_len = lens.to(dev)       # [nchunks] int32
 
     def _topk_ivf(self, q: torch.Tensor, k: int):
-        """IVF probe scoring over cluster-major contiguous slabs.
-
-        IVF is the LOW-LATENCY single/small-query path: it reads only
-        the probed clusters' rows. For large query batches the exact
-        path is faster on this GPU — one GEMM reads the corpus once for
-        ALL queries, while IVF re-reads per query (measured in
-        profiles/RESULTS.md)."""
-        nprobe = min(self.nprobe, self.ivf_nlist)
+        """Batched IVF search: score ALL (query, probe) chunks in one
+        kernel launch (GPU: ivf_score HIP kernel; CPU: one bmm), then a
+        single padded top-k. No per-query host loop (VERDICT.md #7)."""
+        Q = q.shape[0]
+        nprobe = min(self.nprobe, self.ivf_nchunks)
         cscores = q @ self.ivf_centroids.t()
         _, probes = torch.topk(cscores, nprobe, dim=1)   # [Q, nprobe]
-        probes_h = probes.cpu().tolist()
-        vs, is_ = [], []
-        for qi in range(q.shape[0]):
-            parts_v, parts_i = [], []
-            for c in probes_h[qi]:
-                o0 = int(self.ivf_offsets[c])
-                o1 = int(self.ivf_offsets[c + 1])
-                if o0 == o1:
-                    continue
-                parts_v.append(self.corpus_sorted[o0:o1] @ q[qi])
-                parts_i.append(self.ivf_order[o0:o1])
-            scores = torch.cat(parts_v).unsqueeze(0)
-            cand = torch.cat(parts_i).to(scores.device)
-            kk = min(k, cand.numel())
-            v, i = ops.topk(scores.contiguous(), kk)
-            idx = cand[i[0].long()].int()
-            if kk < k:
-                v = torch.nn.functional.pad(v, (0, k - kk),
-                                            value=float("-inf"))
-                idx = torch.nn.functional.pad(idx, (0, k - kk), value=-1)
-            vs.append(v[0])
-            is_.append(idx)
-        return torch.stack(vs), torch.stack(is_)
+        probes = probes.int().contiguous()
+        cap = self.ivf_cap
+        if q.is_cuda and ops.HAVE_HIP:
+            scores = ops.ivf_score(self.ivf_slabs, self.ivf_len,
+                                   probes, q)            # [Q, np, cap]
+        else:
+            sl = self.ivf_slabs[probes.long()]           # [Q, np, cap, D]
+            scores = torch.einsum("qpcd,qd->qpc", sl, q)
+            valid = (torch.arange(cap, device=q.device)[None, None, :]
+                     < self.ivf_len[probes.long()][:, :, None])
+            scores = scores.masked_fill(~valid, float("-inf"))
+        cand = self.ivf_idx[probes.long()].reshape(Q, -1)  # [Q, np*cap]
+        flat = scores.reshape(Q, -1).contiguous()
+        kk = min(k, flat.shape[1])
+        v, i = ops.topk(flat, kk)
+        idx = cand.gather(1, i.long())
+        if kk < k:
+            v = torch.nn.functional.pad(v, (0, k - kk),
+                                        value=float("-inf"))
+            idx = torch.nn.functional.pad(idx, (0, k - kk), value=-1)
+        return v, idx
 
     # ---- search ---------------------------------------------------------
     def transform(self, embeddings: np.ndarray) -> np.ndarray:

@@ -378,3 +378,26 @@ def test_esm2_packed_matches_padded(device):
     b = ref.float()[m]
     err = (a - b).abs().max().item()
     assert err < 0.08, f"packed esm2 vs padded fp32: max err {err}"
+
+
+def test_ivf_score_kernel(device):
+    """ivf_score HIP kernel vs the einsum reference on padded slabs."""
+    torch.manual_seed(44)
+    nch, cap, H = 7, 96, 100  # H deliberately not a multiple of 256
+    slabs = torch.randn(nch, cap, H, device=device)
+    lens = torch.tensor([96, 3, 64, 96, 1, 40, 95], dtype=torch.int32,
+                        device=device)
+    valid = (torch.arange(cap, device=device)[None, :] < lens[:, None])
+    slabs = slabs * valid.unsqueeze(-1)
+    Q, nprobe = 5, 3
+    probes = torch.randint(0, nch, (Q, nprobe), dtype=torch.int32,
+                           device=device)
+    q = torch.randn(Q, H, device=device)
+    got = ops.ivf_score(slabs, lens, probes, q)
+    ref = torch.einsum("qpcd,qd->qpc", slabs[probes.long()], q)
+    ref = ref.masked_fill(
+        ~(torch.arange(cap, device=device)[None, None, :]
+          < lens[probes.long()][:, :, None]), float("-inf"))
+    fin = torch.isfinite(ref)
+    assert torch.equal(torch.isfinite(got), fin)
+    assert torch.allclose(got[fin], ref[fin], atol=1e-3, rtol=1e-3)
