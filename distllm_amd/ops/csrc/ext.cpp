@@ -416,6 +416,7 @@ torch::Tensor ivf_score(torch::Tensor slabs, torch::Tensor lens,
   CHECK_DEV(q); CHECK_CONT(q); CHECK_F32(q);
   const int cap = slabs.size(1), H = slabs.size(2);
   const int Q = probes.size(0), nprobe = probes.size(1);
+  TORCH_CHECK(H % 4 == 0, "slab dim must be padded to a multiple of 4");
   TORCH_CHECK(q.size(0) == Q && q.size(1) == H, "q shape mismatch");
   auto out = torch::empty({Q, nprobe, cap},
                           q.options().dtype(at::kFloat));

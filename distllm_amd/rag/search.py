@@ -229,11 +229,14 @@ class HipFlatIndex:
                             dtype=torch.int32)
         # padded fp32 slabs: exact scores (bf16 slabs measured to
         # scramble near-tie ordering on tight clusters); 2x traffic is
-        # still far below the exact path at small query counts
-        slabs = torch.zeros(nch, cap, self.dim, dtype=torch.float32,
+        # still far below the exact path at small query counts. The dim
+        # is padded to a multiple of 4 so the kernel's 16-byte row
+        # loads stay naturally aligned for any embedding size.
+        Hp = (self.dim + 3) // 4 * 4
+        slabs = torch.zeros(nch, cap, Hp, dtype=torch.float32,
                             device=dev)
         gather = idx_pad.clamp(min=0).long().to(dev)
-        slabs.copy_(
+        slabs[:, :, :self.dim].copy_(
             self.corpus[gather.reshape(-1)].reshape(nch, cap, self.dim))
         # zero the pad rows (the clamp pointed them at corpus[0]): they
         # must not contribute to chunk means nor produce candidates
@@ -241,7 +244,7 @@ class HipFlatIndex:
                  < lens.to(dev)[:, None])
         slabs.mul_(valid.unsqueeze(-1))
         # per-chunk centroid = chunk mean (valid rows only)
-        csum = slabs.float().sum(1)
+        csum = slabs[:, :, :self.dim].sum(1)
         ccent = csum / lens.to(dev).float().clamp(min=1).unsqueeze(1)
 
         self.ivf_nlist = C
@@ -262,12 +265,15 @@ class HipFlatIndex:
         _, probes = torch.topk(cscores, nprobe, dim=1)   # [Q, nprobe]
         probes = probes.int().contiguous()
         cap = self.ivf_cap
+        Hp = self.ivf_slabs.shape[2]
+        qp = (q if Hp == q.shape[1]
+              else torch.nn.functional.pad(q, (0, Hp - q.shape[1])))
         if q.is_cuda and ops.HAVE_HIP:
             scores = ops.ivf_score(self.ivf_slabs, self.ivf_len,
-                                   probes, q)            # [Q, np, cap]
+                                   probes, qp)           # [Q, np, cap]
         else:
             sl = self.ivf_slabs[probes.long()]           # [Q, np, cap, D]
-            scores = torch.einsum("qpcd,qd->qpc", sl, q)
+            scores = torch.einsum("qpcd,qd->qpc", sl, qp)
             valid = (torch.arange(cap, device=q.device)[None, None, :]
                      < self.ivf_len[probes.long()][:, :, None])
             scores = scores.masked_fill(~valid, float("-inf"))

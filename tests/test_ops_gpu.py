@@ -383,7 +383,7 @@ def test_esm2_packed_matches_padded(device):
 def test_ivf_score_kernel(device):
     """ivf_score HIP kernel vs the einsum reference on padded slabs."""
     torch.manual_seed(44)
-    nch, cap, H = 7, 96, 100  # H deliberately not a multiple of 256
+    nch, cap, H = 7, 96, 104  # padded-to-4 dim, not a multiple of 256
     slabs = torch.randn(nch, cap, H, device=device)
     lens = torch.tensor([96, 3, 64, 96, 1, 40, 95], dtype=torch.int32,
                         device=device)
@@ -393,6 +393,12 @@ def test_ivf_score_kernel(device):
     probes = torch.randint(0, nch, (Q, nprobe), dtype=torch.int32,
                            device=device)
     q = torch.randn(Q, H, device=device)
+    # odd-dim corpora are padded to a multiple of 4 by the index build;
+    # the binding rejects unpadded dims
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        ops.ivf_score(slabs[:, :, :101].contiguous(), lens, probes,
+                      q[:, :101].contiguous())
     got = ops.ivf_score(slabs, lens, probes, q)
     ref = torch.einsum("qpcd,qd->qpc", slabs[probes.long()], q)
     ref = ref.masked_fill(
