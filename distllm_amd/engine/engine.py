@@ -127,10 +127,13 @@ class LLMEngine:
         self._next_id = 0
         # hipGraph-captured decode (one graph per pow2 batch bucket):
         # the decode step is ~300 small launches per step for a 32-layer
-        # model — graph replay removes the launch overhead.
-        tp = getattr(model, 'tp', 1)
+        # model — graph replay removes the launch overhead. ROCm 7
+        # supports RCCL collectives inside graph capture, so tp>1
+        # captures the per-layer all-reduces too; if
+        # capture fails on a given stack, _graph_for falls back to
+        # eager decode with a warning.
         self.use_graphs = (use_graphs if use_graphs is not None
-                           else (str(dev).startswith("cuda") and tp == 1))
+                           else str(dev).startswith("cuda"))
         # deterministic sampling RNG shared across TP ranks
         self._rng = torch.Generator(device=dev)
         self._rng.manual_seed(0x5EED)
@@ -234,6 +237,21 @@ class LLMEngine:
         self._sample_and_append(seqs, logits, params, results)
 
     @torch.no_grad()
+    def _graph_for_or_eager(self, n: int):
+        """Graph for the bucket, or None after a failed capture (e.g. a
+        ROCm stack without RCCL-in-graph support for tp>1)."""
+        try:
+            return self._graph_for(n)
+        except RuntimeError as e:
+            import warnings
+
+            warnings.warn(
+                f"[engine] hipGraph decode capture failed ({e}); "
+                "falling back to eager decode", RuntimeWarning)
+            self.use_graphs = False
+            self._graphs.clear()
+            return None
+
     def _graph_for(self, n: int) -> "_GraphRunner":
         bucket = 1
         while bucket < n:
@@ -249,9 +267,12 @@ class LLMEngine:
         tokens = torch.tensor([s.out_ids[-1] for s in seqs], device=dev)
         ctx = torch.tensor([s.length for s in seqs], dtype=torch.int32,
                            device=dev)
+        runner = None
         if self.use_graphs:
+            runner = self._graph_for_or_eager(len(seqs))
+        if runner is not None:
             bt = self._block_table(seqs, width=self.max_blocks_per_seq)
-            logits = self._graph_for(len(seqs)).run(tokens, bt, ctx)
+            logits = runner.run(tokens, bt, ctx)
             logits = logits[: len(seqs)]
         else:
             bt = self._block_table(seqs)
@@ -290,7 +311,8 @@ class LLMEngine:
         if not self.use_graphs:
             return
         for b in batch_sizes or [self.max_batch_size]:
-            self._graph_for(b)
+            if self._graph_for_or_eager(b) is None:
+                return
 
     def _sample_top_p(self, logits: torch.Tensor,
                       params: SamplingParams) -> torch.Tensor:
