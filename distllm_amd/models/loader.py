@@ -85,21 +85,22 @@ def _map_llama(src: dict, model) -> dict:
 
     for i in range(len(model.layers)):
         lp = f"{p}layers.{i}."
-        out[f"layers.{i}.wq"] = rows(src[lp + "self_attn.q_proj.weight"],
-                                     d.num_heads * hd)
-        out[f"layers.{i}.wk"] = rows(src[lp + "self_attn.k_proj.weight"],
-                                     d.kv_heads * hd)
-        out[f"layers.{i}.wv"] = rows(src[lp + "self_attn.v_proj.weight"],
-                                     d.kv_heads * hd)
+        # fused column-parallel parameters: each part is TP-sliced
+        # independently, then concatenated along the output rows
+        out[f"layers.{i}.wqkv"] = torch.cat([
+            rows(src[lp + "self_attn.q_proj.weight"], d.num_heads * hd),
+            rows(src[lp + "self_attn.k_proj.weight"], d.kv_heads * hd),
+            rows(src[lp + "self_attn.v_proj.weight"], d.kv_heads * hd),
+        ], 0)
         out[f"layers.{i}.wo"] = cols(src[lp + "self_attn.o_proj.weight"],
                                      d.num_heads * hd)
         out[f"layers.{i}.in_norm"] = src[lp + "input_layernorm.weight"]
         out[f"layers.{i}.post_norm"] = src[
             lp + "post_attention_layernorm.weight"]
-        out[f"layers.{i}.wgate"] = rows(src[lp + "mlp.gate_proj.weight"],
-                                        d.intermediate_size)
-        out[f"layers.{i}.wup"] = rows(src[lp + "mlp.up_proj.weight"],
-                                      d.intermediate_size)
+        out[f"layers.{i}.wgup"] = torch.cat([
+            rows(src[lp + "mlp.gate_proj.weight"], d.intermediate_size),
+            rows(src[lp + "mlp.up_proj.weight"], d.intermediate_size),
+        ], 0)
         out[f"layers.{i}.wdown"] = cols(src[lp + "mlp.down_proj.weight"],
                                         d.intermediate_size)
     return out

@@ -24,6 +24,13 @@ from .config import ModelDims
 
 
 class DecoderLayer(nn.Module):
+    """Column-parallel fused qkv & gate|up, row-parallel wo & wdown
+    (Megatron-style sharding; all-reduce after wo and wdown). The q/k/v
+    and gate/up projections live in ONE parameter each so the decode
+    step issues ONE GEMM for qkv and ONE for the MLP up-path
+    (VERDICT.md #5: 7 -> 4 weight streams per layer); wq/wk/... remain
+    available as row-slice views for checkpoint export and tests."""
+
     def __init__(self, d: ModelDims, tp: int = 1):
         super().__init__()
         H = d.hidden_size
@@ -32,17 +39,34 @@ class DecoderLayer(nn.Module):
         nh = d.num_heads // tp
         hd = d.head_dim
         inter = d.intermediate_size // tp
-        # column-parallel q/k/v & gate/up, row-parallel wo & wdown
-        # (Megatron-style sharding; all-reduce after wo and wdown)
-        self.wq = nn.Parameter(torch.empty(nh * hd, H))
-        self.wk = nn.Parameter(torch.empty(kvh * hd, H))
-        self.wv = nn.Parameter(torch.empty(kvh * hd, H))
+        self._nq, self._nkv = nh * hd, kvh * hd
+        self._inter = inter
+        self.wqkv = nn.Parameter(torch.empty((nh + 2 * kvh) * hd, H))
         self.wo = nn.Parameter(torch.empty(H, nh * hd))
         self.in_norm = nn.Parameter(torch.empty(H))
         self.post_norm = nn.Parameter(torch.empty(H))
-        self.wgate = nn.Parameter(torch.empty(inter, H))
-        self.wup = nn.Parameter(torch.empty(inter, H))
+        self.wgup = nn.Parameter(torch.empty(2 * inter, H))
         self.wdown = nn.Parameter(torch.empty(H, inter))
+
+    @property
+    def wq(self):
+        return self.wqkv[: self._nq]
+
+    @property
+    def wk(self):
+        return self.wqkv[self._nq: self._nq + self._nkv]
+
+    @property
+    def wv(self):
+        return self.wqkv[self._nq + self._nkv:]
+
+    @property
+    def wgate(self):
+        return self.wgup[: self._inter]
+
+    @property
+    def wup(self):
+        return self.wgup[self._inter:]
 
 
 class DecoderModel(nn.Module):
@@ -138,11 +162,12 @@ class DecoderModel(nn.Module):
         B, S, _ = h.shape
         nh, kvh, hd = self.local_heads, self.local_kv_heads, d.head_dim
         cs, sn = self._rope(h.device)
-        q = ops.linear(h, layer.wq).view(B, S, nh, hd)
-        k = ops.linear(h, layer.wk).view(B, S, kvh, hd)
-        v = ops.linear(h, layer.wv).view(B, S, kvh, hd)
-        q = ops.rope_apply(q, cs, sn, pos)
-        k = ops.rope_apply(k, cs, sn, pos)
+        # ONE fused qkv GEMM, ONE rope launch (q|k roped, v copied)
+        qkv = ops.linear(h, layer.wqkv).view(B, S, nh + 2 * kvh, hd)
+        qkv = ops.rope_apply(qkv, cs, sn, pos, rope_heads=nh + kvh)
+        q = qkv[:, :, :nh]
+        k = qkv[:, :, nh: nh + kvh]
+        v = qkv[:, :, nh + kvh:]
         if kv_out is not None:
             kv_out.append((k, v))
         ctx = ops.attn_fwd(q, k, v, mask=mask_f, causal=True,
@@ -152,10 +177,8 @@ class DecoderModel(nn.Module):
         # fused: x' = x + attn ; h2 = rms_norm(x')
         h2, x = ops.norm_fwd_sum(attn, x, layer.post_norm,
                                  eps=d.rms_norm_eps, rms=True)
-        gate = ops.linear(h2, layer.wgate)
-        up = ops.linear(h2, layer.wup)
         down = self._allreduce(
-            ops.linear(ops.swiglu(gate, up), layer.wdown))
+            ops.linear(ops.gateup_swiglu(h2, layer.wgup), layer.wdown))
         return down, x
 
     @torch.no_grad()
@@ -217,24 +240,25 @@ class DecoderModel(nn.Module):
         slot = pos
         h = ops.rms_norm(x, self.layers[0].in_norm, eps=d.rms_norm_eps)
         for li, layer in enumerate(self.layers):
-            q = ops.linear(h, layer.wq).view(B, 1, nh, hd)
-            k = ops.linear(h, layer.wk).view(B, 1, kvh, hd)
-            v = ops.linear(h, layer.wv).view(B, 1, kvh, hd)
-            q = ops.rope_apply(q, cs, sn, pos)
-            k = ops.rope_apply(k, cs, sn, pos)
-            ops.cache_put(k[:, 0], v[:, 0], kcaches[li], vcaches[li],
+            # ONE fused qkv GEMM + ONE rope (q|k roped, v copied):
+            # 5 launches -> 2 on the per-token critical path
+            qkv = ops.linear(h, layer.wqkv).view(B, 1, nh + 2 * kvh, hd)
+            qkv = ops.rope_apply(qkv, cs, sn, pos, rope_heads=nh + kvh)
+            q = qkv[:, 0, :nh]
+            k = qkv[:, 0, nh: nh + kvh].contiguous()
+            v = qkv[:, 0, nh + kvh:].contiguous()
+            ops.cache_put(k, v, kcaches[li], vcaches[li],
                           block_tables, slot)
-            ctx = ops.paged_attn(q[:, 0].contiguous(), kcaches[li],
+            ctx = ops.paged_attn(q.contiguous(), kcaches[li],
                                  vcaches[li], block_tables, ctx_lens,
                                  scale=1.0 / math.sqrt(hd))
             attn = self._allreduce(
                 ops.linear(ctx.reshape(B, 1, nh * hd), layer.wo))
             h2, x = ops.norm_fwd_sum(attn, x, layer.post_norm,
                                      eps=d.rms_norm_eps, rms=True)
-            gate = ops.linear(h2, layer.wgate)
-            up = ops.linear(h2, layer.wup)
             down = self._allreduce(
-                ops.linear(ops.swiglu(gate, up), layer.wdown))
+                ops.linear(ops.gateup_swiglu(h2, layer.wgup),
+                           layer.wdown))
             next_w = (self.layers[li + 1].in_norm
                       if li + 1 < len(self.layers) else self.final_norm)
             h, x = ops.norm_fwd_sum(down, x, next_w,

@@ -372,11 +372,37 @@ def attn_varlen(Q, K, V, cu_seqlens, block_map=None, scale=None):
     return torch_ref.attn_varlen(Q, K, V, cu_seqlens, scale)
 
 
-def rope_apply(x, cs, sn, pos):
+def rope_apply(x, cs, sn, pos, rope_heads: int = -1):
+    """Heads >= rope_heads (when >= 0) pass through unchanged — lets a
+    fused qkv tensor rope its q|k heads and copy v in ONE launch."""
     if x.is_cuda:
         _require_hip()
-        return _hip_ops.rope_apply(x.contiguous(), cs, sn, pos.int())
-    return torch_ref.rope_apply(x, cs, sn, pos)
+        return _hip_ops.rope_apply(x.contiguous(), cs, sn, pos.int(),
+                                   rope_heads)
+    return torch_ref.rope_apply(x, cs, sn, pos, rope_heads)
+
+
+def gateup_swiglu(x, wgup, bias=None):
+    """silu(x @ Wg^T) * (x @ Wu^T) with Wg|Wu packed as wgup [2I, K].
+    Decode path (M <= 256): ONE split-K GEMM with the swiglu fused into
+    its reduce; otherwise one big GEMM + packed swiglu."""
+    orig_shape = x.shape
+    x2 = x.reshape(-1, orig_shape[-1])
+    M, K = x2.shape
+    I = wgup.shape[0] // 2
+    if x.is_cuda:
+        _require_hip()
+        if M <= 256 and wgup.shape[0] % 128 == 0 and x2.is_contiguous():
+            sk = _decode_splitk(M, wgup.shape[0], K)
+            if sk > 1:
+                out = _hip_ops.decode_gemm(x2, wgup, bias, sk, 1)
+                return out.reshape(*orig_shape[:-1], I)
+    gu = linear(x2, wgup, bias)
+    if gu.is_cuda:
+        out = _hip_ops.swiglu_packed(gu.contiguous())
+    else:
+        out = swiglu(gu[:, :I], gu[:, I:])
+    return out.reshape(*orig_shape[:-1], I)
 
 
 rope_tables = torch_ref.rope_tables

@@ -56,6 +56,43 @@ swiglu_kernel(const unsigned short* __restrict__ gate,
   }
 }
 
+
+
+// Packed swiglu: gu [R, 2I] (gate cols then up cols) -> out [R, I].
+// Lets the fused gate|up GEMM output feed the MLP without slicing the
+// halves into contiguous copies.
+__global__ void __launch_bounds__(256)
+swiglu_packed_kernel(const unsigned short* __restrict__ gu,
+                     unsigned short* __restrict__ out, long R, int I) {
+  const long n_vec = (R * (long)I) >> 3;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const int iv = I >> 3;
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < n_vec;
+       v += stride) {
+    const long r = v / iv;
+    const int c = (int)(v % iv);
+    const unsigned short* row = gu + r * (2L * I);
+    s16x8 gv = *reinterpret_cast<const s16x8*>(row + c * 8);
+    s16x8 uv = *reinterpret_cast<const s16x8*>(row + I + c * 8);
+    float g[8], u[8];
+    bf8_to_f32(gv, g);
+    bf8_to_f32(uv, u);
+    s16x8 ov;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) ov[i] = (short)f2bf(silu(g[i]) * u[i]);
+    *reinterpret_cast<s16x8*>(out + v * 8) = ov;
+  }
+}
+
+extern "C" void launch_swiglu_packed(const void* gu, void* out, long R,
+                                     int I, hipStream_t stream) {
+  const long n_vec = (R * (long)I) >> 3;
+  int blocks = (int)min((n_vec + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(swiglu_packed_kernel, dim3(blocks), dim3(256), 0,
+                     stream, (const unsigned short*)gu,
+                     (unsigned short*)out, R, I);
+}
+
 // Fused bias-add (no activation), used for attention output projections
 // where hipBLASLt's bias epilogue is not engaged.
 __global__ void __launch_bounds__(256)

@@ -27,6 +27,7 @@ void launch_norm_fwd(const void*, const void*, const void*, const void*,
 void launch_bias_gelu(const void*, const void*, void*, long, int, bool,
                       hipStream_t);
 void launch_swiglu(const void*, const void*, void*, long, hipStream_t);
+void launch_swiglu_packed(const void*, void*, long, int, hipStream_t);
 void launch_bias_add(const void*, const void*, void*, long, int, hipStream_t);
 void launch_mean_pool(const void*, const int*, float*, int, int, int, bool,
                       hipStream_t);
@@ -41,7 +42,7 @@ void launch_gemm8_bf16(const void*, const void*, const void*, void*, int, int,
 void launch_gemm16_bf16(const void*, const void*, const void*, void*, int,
                         int, int, int, hipStream_t);
 void launch_decode_gemm(const void*, const void*, const void*, void*, void*,
-                        int, int, int, int, hipStream_t);
+                        int, int, int, int, int, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, const float*,
                      void*, int, int, int, int, int, int, bool, float, long,
                      long, long, long, hipStream_t);
@@ -49,7 +50,7 @@ void launch_attn_varlen(const void*, const void*, const void*, void*,
                         const int*, const int*, const int*, int, int, int,
                         int, int, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, const int*,
-                 long, int, int, hipStream_t);
+                 long, int, int, int, hipStream_t);
 void launch_topk_chunk(const float*, const int*, float*, int*, int, long, int,
                        int, hipStream_t);
 void launch_hamming_score(const void*, const void*, float*, int, long, int,
@@ -95,6 +96,18 @@ torch::Tensor bias_gelu(torch::Tensor x, c10::optional<torch::Tensor> bias,
   TORCH_CHECK(N % 8 == 0, "last dim must be mult of 8");
   launch_bias_gelu(x.data_ptr(), bias ? bias->data_ptr() : nullptr,
                    out.data_ptr(), x.numel(), N, erf, cur_stream());
+  return out;
+}
+
+torch::Tensor swiglu_packed(torch::Tensor gu) {
+  CHECK_DEV(gu); CHECK_CONT(gu); CHECK_BF16(gu);
+  const int I = gu.size(-1) / 2;
+  TORCH_CHECK(I % 8 == 0, "I must be a multiple of 8");
+  const long R = gu.numel() / (2L * I);
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gu.options());
+  launch_swiglu_packed(gu.data_ptr(), out.data_ptr(), R, I, cur_stream());
   return out;
 }
 
@@ -204,7 +217,9 @@ torch::Tensor gemm16_bf16(torch::Tensor A, torch::Tensor B,
 
 torch::Tensor decode_gemm(torch::Tensor X, torch::Tensor W,
                           c10::optional<torch::Tensor> bias,
-                          int64_t splitk) {
+                          int64_t splitk, int64_t act = 0) {
+  // act 1 = packed swiglu in the reduce: W is [2I, K] (gate rows then
+  // up rows), output is [M, I]; requires splitk > 1.
   CHECK_DEV(X); CHECK_CONT(X); CHECK_BF16(X);
   CHECK_DEV(W); CHECK_CONT(W); CHECK_BF16(W);
   const int M = X.size(0), K = X.size(1), N = W.size(0);
@@ -212,8 +227,10 @@ torch::Tensor decode_gemm(torch::Tensor X, torch::Tensor W,
   TORCH_CHECK(M <= 256, "decode_gemm is for M <= 256");
   TORCH_CHECK(N % 64 == 0, "N % 64");
   TORCH_CHECK(K % (64 * splitk) == 0, "K % (64*splitk)");
+  TORCH_CHECK(act == 0 || (splitk > 1 && N % 128 == 0),
+              "swiglu fusion needs splitk > 1 and even I");
   const int MP = M <= 128 ? 128 : 256;
-  auto out = torch::empty({M, N}, X.options());
+  auto out = torch::empty({M, act == 1 ? N / 2 : N}, X.options());
   torch::Tensor slabs;
   void* slabp = nullptr;
   if (splitk > 1) {
@@ -223,7 +240,7 @@ torch::Tensor decode_gemm(torch::Tensor X, torch::Tensor W,
   }
   launch_decode_gemm(X.data_ptr(), W.data_ptr(),
                      bias ? bias->data_ptr() : nullptr, out.data_ptr(),
-                     slabp, M, N, K, (int)splitk, cur_stream());
+                     slabp, M, N, K, (int)splitk, (int)act, cur_stream());
   return out;
 }
 
@@ -289,7 +306,7 @@ torch::Tensor attn_varlen(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
 
 // ---- rope ------------------------------------------------------------------
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
-                         torch::Tensor pos) {
+                         torch::Tensor pos, int64_t rope_heads = -1) {
   CHECK_DEV(x); CHECK_CONT(x); CHECK_BF16(x);
   CHECK_F32(cs); CHECK_F32(sn); CHECK_I32(pos);
   const int D = x.size(-1);
@@ -298,7 +315,7 @@ torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
   auto out = torch::empty_like(x);
   launch_rope(x.data_ptr(), out.data_ptr(), cs.data_ptr<float>(),
               sn.data_ptr<float>(), pos.data_ptr<int>(), n_rows, Hx, D,
-              cur_stream());
+              rope_heads < 0 ? Hx : (int)rope_heads, cur_stream());
   return out;
 }
 
@@ -433,6 +450,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "fused residual+LayerNorm/RMSNorm");
   m.def("bias_gelu", &bias_gelu);
   m.def("swiglu", &swiglu);
+  m.def("swiglu_packed", &swiglu_packed);
   m.def("bias_add", &bias_add);
   m.def("mean_pool", &mean_pool);
   m.def("last_token_pool", &last_token_pool);
@@ -441,10 +459,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16", &gemm_bf16);
   m.def("gemm8_bf16", &gemm8_bf16);
   m.def("gemm16_bf16", &gemm16_bf16);
-  m.def("decode_gemm", &decode_gemm);
+  m.def("decode_gemm", &decode_gemm, py::arg("X"), py::arg("W"),
+        py::arg("bias"), py::arg("splitk"), py::arg("act") = 0);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_varlen", &attn_varlen);
-  m.def("rope_apply", &rope_apply);
+  m.def("rope_apply", &rope_apply, py::arg("x"), py::arg("cs"),
+        py::arg("sn"), py::arg("pos"), py::arg("rope_heads") = -1);
   m.def("topk_chunked", &topk_chunked);
   m.def("hamming_score", &hamming_score);
   m.def("ivf_score", &ivf_score);

@@ -16,7 +16,10 @@ rope_kernel(const unsigned short* __restrict__ x,
             const float* __restrict__ sn,   // [max_pos, D/2] sin
             const int* __restrict__ pos,    // [B*S] absolute positions
             long n_rows,                    // B*S*Hx
-            int Hx, int D) {
+            int Hx, int D, int rope_heads) {
+  // heads >= rope_heads pass through unchanged: lets ONE launch rope
+  // the q|k part of a fused qkv tensor while copying v (decode-path
+  // launch fusion)
   const int half = D / 2;
   const long stride = (long)gridDim.x * blockDim.x;
   const long total = n_rows * half;
@@ -25,6 +28,12 @@ rope_kernel(const unsigned short* __restrict__ x,
     const long row = i / half;
     const int d = (int)(i % half);
     const long tok = row / Hx;
+    if ((int)(row - tok * Hx) >= rope_heads) {
+      const unsigned short* xr = x + row * D;
+      out[row * D + d] = xr[d];
+      out[row * D + d + half] = xr[d + half];
+      continue;
+    }
     const int p = pos[tok];
     const float c = cs[(long)p * half + d];
     const float s = sn[(long)p * half + d];
@@ -38,10 +47,11 @@ rope_kernel(const unsigned short* __restrict__ x,
 
 extern "C" void launch_rope(const void* x, void* out, const float* cs,
                             const float* sn, const int* pos, long n_rows,
-                            int Hx, int D, hipStream_t stream) {
+                            int Hx, int D, int rope_heads,
+                            hipStream_t stream) {
   const long total = n_rows * (D / 2);
   int blocks = (int)min((total + 255) / 256, (long)2048);
   hipLaunchKernelGGL(rope_kernel, dim3(blocks), dim3(256), 0, stream,
                      (const unsigned short*)x, (unsigned short*)out, cs, sn,
-                     pos, n_rows, Hx, D);
+                     pos, n_rows, Hx, D, rope_heads);
 }

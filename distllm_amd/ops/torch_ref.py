@@ -191,8 +191,9 @@ def rope_tables(max_pos: int, D: int, theta: float = 10000.0, device="cpu"):
 
 
 def rope_apply(x: torch.Tensor, cs: torch.Tensor, sn: torch.Tensor,
-               pos: torch.Tensor):
-    """x [..., S, Hx, D] flattened rows [B,S,Hx,D]; pos int32 [B*S]."""
+               pos: torch.Tensor, rope_heads: int = -1):
+    """x [..., S, Hx, D] flattened rows [B,S,Hx,D]; pos int32 [B*S].
+    Heads >= rope_heads (when >= 0) pass through unchanged."""
     shape = x.shape
     D = shape[-1]
     Hx = shape[-2]
@@ -202,6 +203,8 @@ def rope_apply(x: torch.Tensor, cs: torch.Tensor, sn: torch.Tensor,
     s = sn[p].unsqueeze(1)
     x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
     out = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+    if 0 <= rope_heads < Hx:
+        out[:, rope_heads:] = xf[:, rope_heads:]
     return out.reshape(shape).to(x.dtype)
 
 

@@ -57,24 +57,23 @@ def _bert_to_hf(model) -> dict:
 
 
 def _llama_to_hf(model) -> dict:
-    p = dict(model.named_parameters())
     out = {
-        "model.embed_tokens.weight": p["tok_emb"],
-        "model.norm.weight": p["final_norm"],
-        "lm_head.weight": p["lm_head"],
+        "model.embed_tokens.weight": model.tok_emb,
+        "model.norm.weight": model.final_norm,
+        "lm_head.weight": model.lm_head,
     }
-    for i in range(len(model.layers)):
+    for i, lyr in enumerate(model.layers):
         lp = f"model.layers.{i}."
-        out[lp + "self_attn.q_proj.weight"] = p[f"layers.{i}.wq"]
-        out[lp + "self_attn.k_proj.weight"] = p[f"layers.{i}.wk"]
-        out[lp + "self_attn.v_proj.weight"] = p[f"layers.{i}.wv"]
-        out[lp + "self_attn.o_proj.weight"] = p[f"layers.{i}.wo"]
-        out[lp + "input_layernorm.weight"] = p[f"layers.{i}.in_norm"]
-        out[lp + "post_attention_layernorm.weight"] = \
-            p[f"layers.{i}.post_norm"]
-        out[lp + "mlp.gate_proj.weight"] = p[f"layers.{i}.wgate"]
-        out[lp + "mlp.up_proj.weight"] = p[f"layers.{i}.wup"]
-        out[lp + "mlp.down_proj.weight"] = p[f"layers.{i}.wdown"]
+        # wq/wk/wv/wgate/wup are row-slice views of the fused params
+        out[lp + "self_attn.q_proj.weight"] = lyr.wq
+        out[lp + "self_attn.k_proj.weight"] = lyr.wk
+        out[lp + "self_attn.v_proj.weight"] = lyr.wv
+        out[lp + "self_attn.o_proj.weight"] = lyr.wo
+        out[lp + "input_layernorm.weight"] = lyr.in_norm
+        out[lp + "post_attention_layernorm.weight"] = lyr.post_norm
+        out[lp + "mlp.gate_proj.weight"] = lyr.wgate
+        out[lp + "mlp.up_proj.weight"] = lyr.wup
+        out[lp + "mlp.down_proj.weight"] = lyr.wdown
     return out
 
 

@@ -407,3 +407,53 @@ def test_ivf_score_kernel(device):
     fin = torch.isfinite(ref)
     assert torch.equal(torch.isfinite(got), fin)
     assert torch.allclose(got[fin], ref[fin], atol=1e-3, rtol=1e-3)
+
+
+def test_rope_partial_heads(device):
+    torch.manual_seed(50)
+    B, S, Hx, D = 2, 5, 6, 64
+    x = torch.randn(B, S, Hx, D, device=device, dtype=torch.bfloat16)
+    cs, sn = torch_ref.rope_tables(32, D, device=device)
+    pos = torch.arange(S, device=device).repeat(B).int()
+    out = ops.rope_apply(x, cs, sn, pos, rope_heads=4)
+    ref = torch_ref.rope_apply(x, cs, sn, pos, rope_heads=4)
+    _bf16_close(out, ref, atol=2e-2, rtol=2e-2)
+    # heads >= rope_heads must be bit-identical pass-through
+    assert torch.equal(out[:, :, 4:], x[:, :, 4:])
+
+
+def test_swiglu_packed(device):
+    torch.manual_seed(51)
+    gu = torch.randn(7, 256, device=device, dtype=torch.bfloat16)
+    from distllm_amd.ops import _hip_ops
+
+    out = _hip_ops.swiglu_packed(gu)
+    ref = torch_ref.swiglu(gu[:, :128], gu[:, 128:])
+    _bf16_close(out, ref, atol=2e-2, rtol=2e-2)
+
+
+def test_decode_gemm_swiglu_fused(device):
+    """decode_gemm act=1: packed gate|up weights, swiglu in the
+    split-K reduce."""
+    torch.manual_seed(52)
+    from distllm_amd.ops import _hip_ops
+
+    M, I, K = 3, 256, 512
+    x = torch.randn(M, K, device=device, dtype=torch.bfloat16) * 0.3
+    w = torch.randn(2 * I, K, device=device, dtype=torch.bfloat16) * 0.3
+    out = _hip_ops.decode_gemm(x, w, None, 4, 1)
+    gu = torch_ref.gemm(x, w)
+    ref = torch_ref.swiglu(gu[:, :I], gu[:, I:])
+    assert out.shape == (M, I)
+    _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gateup_swiglu_large_m(device):
+    """The prefill-path (M > 256) packed gate|up route."""
+    torch.manual_seed(53)
+    x = torch.randn(300, 256, device=device, dtype=torch.bfloat16) * 0.3
+    w = torch.randn(512, 256, device=device, dtype=torch.bfloat16) * 0.3
+    out = ops.gateup_swiglu(x, w)
+    gu = torch_ref.gemm(x, w)
+    ref = torch_ref.swiglu(gu[:, :256], gu[:, 256:])
+    _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
