@@ -284,6 +284,21 @@ class LLMEngine:
         dev = self.device
         if params.top_p < 1.0:
             next_ids = self._sample_top_p(logits, params)
+        elif len(seqs) <= 16 and params.temperature > 0 \
+                and str(dev).startswith("cuda"):
+            # small-batch Gumbel-max draw: the CDF-walk kernel is one
+            # block per row (latency-bound, ~170us at B=1 over a 128k
+            # vocab); argmax over logits/T + Gumbel noise is exact for
+            # categorical sampling and fully parallel. Seeded engine
+            # RNG keeps TP ranks identical.
+            lf = logits.float() / max(params.temperature, 1e-5)
+            if params.min_p > 0:
+                floor = (lf.max(dim=-1, keepdim=True).values
+                         + math.log(params.min_p))
+                lf = lf.masked_fill(lf < floor, float("-inf"))
+            u = torch.rand(lf.shape, device=dev, generator=self._rng)
+            g = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))
+            next_ids = (lf + g).argmax(dim=-1).int()
         else:
             rand = torch.rand(len(seqs), device=dev,
                               generator=self._rng)

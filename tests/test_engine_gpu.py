@@ -113,3 +113,19 @@ def test_engine_loop_stress_gpu(model):
     for (p, mt), o in zip(reqs, outs):
         assert len(o) == mt, f"expected {mt} tokens, got {len(o)}"
         assert all(0 <= t < model.dims.vocab_size for t in o)
+
+
+def test_small_batch_gumbel_sampling(model):
+    """B<=16 sampling goes through the Gumbel-max path: it must stay
+    deterministic for a fixed seed and respect min_p (never emit
+    tokens whose probability is filtered out)."""
+    engine = LLMEngine(model, max_batch_size=2, max_model_len=64,
+                       num_blocks=64)
+    params = SamplingParams(temperature=0.7, min_p=0.3, max_tokens=6,
+                            stop_token_ids=())
+    out1 = engine.generate([[5, 9, 100]], params)[0]
+    engine2 = LLMEngine(model, max_batch_size=2, max_model_len=64,
+                        num_blocks=64)
+    out2 = engine2.generate([[5, 9, 100]], params)[0]
+    assert out1 == out2  # same seeded RNG -> same draws
+    assert len(out1) == 6
