@@ -68,6 +68,46 @@ class ConversationPromptTemplate:
         return [r.strip() for r in responses]
 
 
+def translate_target_config(gen: dict[str, Any]) -> dict[str, Any]:
+    """Accept the reference chat_argoproxy `_target_` class-dispatch
+    YAMLs (reference distllm/chat_argoproxy.py:510-549) by translating
+    the three generator config classes onto our generator names:
+
+      VLLMGeneratorConfig    {server, port, api_key, model, ...}
+      ArgoGeneratorConfig    {model, base_url, api_key, ...}
+      OpenAIAPIGeneratorConfig {model, api_key, base_url?, ...}
+
+    all map to the OpenAI-compatible HTTP client; dicts without
+    `_target_` pass through unchanged."""
+    if "_target_" not in gen:
+        return gen
+    gen = dict(gen)
+    target = gen.pop("_target_")
+    common = {
+        k: gen[k] for k in ("model", "temperature", "max_tokens")
+        if k in gen
+    }
+    if target == "VLLMGeneratorConfig":
+        out = {"name": "openai_http",
+               "base_url": f"http://{gen['server']}:{gen['port']}/v1",
+               "api_key": gen.get("api_key", ""), **common}
+    elif target == "ArgoGeneratorConfig":
+        out = {"name": "openai_http",
+               "base_url": gen.get("base_url", "http://localhost:56267"),
+               "api_key": gen.get("api_key", ""), **common}
+    elif target == "OpenAIAPIGeneratorConfig":
+        out = {"name": "openai_http",
+               "base_url": gen.get("base_url")
+               or "https://api.openai.com/v1",
+               "api_key": gen.get("api_key", ""), **common}
+    else:
+        raise ValueError(
+            f"Unknown generator config class: {target}. Available: "
+            "['VLLMGeneratorConfig', 'ArgoGeneratorConfig', "
+            "'OpenAIAPIGeneratorConfig']")
+    return out
+
+
 class ChatAppConfig(BaseConfig):
     generator_config: dict[str, Any]
     retriever_config: RetrieverConfig | None = None
@@ -82,7 +122,11 @@ class ChatAppConfig(BaseConfig):
 
         with open(path) as fh:
             data = yaml.safe_load(fh)
-        return cls(**substitute_env(data or {}))
+        data = substitute_env(data or {})
+        if isinstance(data.get("generator_config"), dict):
+            data["generator_config"] = translate_target_config(
+                data["generator_config"])
+        return cls(**data)
 
 
 class ChatSession:

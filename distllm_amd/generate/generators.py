@@ -172,6 +172,10 @@ class OpenAICompatGeneratorConfig(BaseConfig):
     api_key_env: str = Field(
         "OPENAI_API_KEY", description="env var holding the key, if any"
     )
+    api_key: str = Field(
+        "", description="literal key (reference argoproxy-style configs "
+        "pass the key inline; takes precedence over api_key_env)"
+    )
     temperature: float = 0.5
     max_tokens: int = 2000
     timeout: float = 120.0
@@ -184,7 +188,7 @@ class OpenAICompatGenerator:
 
         self.config = config
         headers = {}
-        key = os.environ.get(config.api_key_env, "")
+        key = config.api_key or os.environ.get(config.api_key_env, "")
         if key:
             headers["Authorization"] = f"Bearer {key}"
         self._client = httpx.Client(base_url=config.base_url,

@@ -227,3 +227,46 @@ def test_generate_worker_skips_empty_results(tmp_path, monkeypatch):
                     {"name": "native", "model": "llama-tiny",
                      "max_tokens": 2, "max_batch_size": 2})
     assert list(out.iterdir()) == []
+
+
+def test_chat_target_dispatch(tmp_path, monkeypatch):
+    """Reference chat_argoproxy `_target_` YAMLs load drop-in
+    (chat_argoproxy.py:510-549 class-dispatch surface)."""
+    import yaml
+
+    from distllm_amd.chat import ChatAppConfig, translate_target_config
+
+    monkeypatch.setenv("ARGO_KEY", "sekret")
+    cfg = {
+        "generator_config": {
+            "_target_": "ArgoGeneratorConfig",
+            "model": "argo:gpt-4o",
+            "base_url": "http://localhost:56267",
+            "api_key": "${env:ARGO_KEY}",
+            "temperature": 0.0,
+            "max_tokens": 128,
+        },
+        "system_prompt": "hi",
+    }
+    p = tmp_path / "chat.yaml"
+    p.write_text(yaml.safe_dump(cfg))
+    app = ChatAppConfig.from_yaml(p)
+    g = app.generator_config
+    assert g["name"] == "openai_http"
+    assert g["base_url"] == "http://localhost:56267"
+    assert g["api_key"] == "sekret"  # ${env:VAR} resolved
+    assert g["model"] == "argo:gpt-4o" and g["max_tokens"] == 128
+
+    v = translate_target_config(
+        {"_target_": "VLLMGeneratorConfig", "server": "rbdgx1",
+         "port": 8000, "api_key": "CELS", "model": "m"})
+    assert v["base_url"] == "http://rbdgx1:8000/v1"
+
+    o = translate_target_config(
+        {"_target_": "OpenAIAPIGeneratorConfig", "model": "gpt-4.1",
+         "api_key": "k"})
+    assert o["base_url"] == "https://api.openai.com/v1"
+
+    import pytest as _p
+    with _p.raises(ValueError, match="Unknown generator config"):
+        translate_target_config({"_target_": "NopeConfig"})
