@@ -457,3 +457,23 @@ def test_gateup_swiglu_large_m(device):
     gu = torch_ref.gemm(x, w)
     ref = torch_ref.swiglu(gu[:, :256], gu[:, 256:])
     _bf16_close(out, ref, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("rows", [1, 3, 8])
+def test_norm_small_batch_variant(device, rows):
+    """n_rows <= 8 dispatches the block-per-row kernel; numerics must
+    match the reference for both LN and RMS, with residual+sum."""
+    torch.manual_seed(60 + rows)
+    H = 4096
+    x = torch.randn(rows, H, device=device, dtype=torch.bfloat16)
+    r = torch.randn(rows, H, device=device, dtype=torch.bfloat16)
+    w = torch.randn(H, device=device, dtype=torch.bfloat16)
+    b = torch.randn(H, device=device, dtype=torch.bfloat16)
+    out = ops.layer_norm(x, w, b, eps=1e-5, res=r)
+    ref = torch_ref.norm_fwd(x, r, w, b, 1e-5, False)
+    _bf16_close(out, ref, atol=3e-2, rtol=3e-2)
+    o2, s2 = ops.norm_fwd_sum(x, r, w, eps=1e-5, rms=True)
+    ref2, refs2 = torch_ref.norm_fwd(x, r, w, None, 1e-5, True,
+                                     return_sum=True)
+    _bf16_close(o2, ref2, atol=3e-2, rtol=3e-2)
+    _bf16_close(s2, refs2, atol=3e-2, rtol=3e-2)
