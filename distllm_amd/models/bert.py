@@ -110,8 +110,15 @@ class BertEncoder(nn.Module):
         attention_mask: torch.Tensor,
         token_type_ids: torch.Tensor | None = None,
     ) -> torch.Tensor:
+        import os
+
         ids = input_ids.long()
         B, S = ids.shape
+        if (ids.is_cuda and ops.HAVE_HIP and attention_mask is not None
+                and not os.environ.get("DISTLLM_NO_VARLEN")
+                and token_type_ids is None
+                and not bool(attention_mask.all())):
+            return self._forward_packed(ids, attention_mask)
         pos = torch.arange(S, device=ids.device)
         x = self.word_emb[ids] + self.pos_emb[pos][None, :, :]
         if token_type_ids is not None:
@@ -129,3 +136,44 @@ class BertEncoder(nn.Module):
         for layer in self.layers:
             x = layer(x, mask_f)
         return x
+
+    @torch.no_grad()
+    def _forward_packed(self, ids, attention_mask):
+        """Varlen path for padded text batches (jsonl_chunk buffers are
+        length-dispersed): valid tokens packed to [T, H], GEMMs/norms on
+        packed rows, attention through the cu_seqlens flash kernel
+        (same design as the ESM packed path). Returns
+        padded [B, S, H] with zeros at padding."""
+        d = self.dims
+        B, S = ids.shape
+        maskb = attention_mask.bool()
+        lens = attention_mask.int().sum(1)
+        cu = torch.zeros(B + 1, dtype=torch.int32, device=ids.device)
+        cu[1:] = lens.cumsum(0)
+        pos = (attention_mask.int().cumsum(1) - 1).clamp(min=0)[maskb]
+        bmap = ops.varlen_block_map(cu)
+        x = (self.word_emb[ids[maskb]] + self.pos_emb[pos.long()]
+             + self.type_emb[0])
+        x = ops.layer_norm(x, self.emb_ln_w, self.emb_ln_b,
+                           eps=d.layer_norm_eps)
+        nh, hd = d.num_heads, d.head_dim
+        H = d.hidden_size
+        T = x.shape[0]
+        scale = 1.0 / math.sqrt(hd)
+        for layer in self.layers:
+            qkv = ops.linear(x, layer.wqkv, layer.bqkv)
+            q = qkv[..., :H].view(T, nh, hd).contiguous()
+            k = qkv[..., H: 2 * H].view(T, nh, hd).contiguous()
+            v = qkv[..., 2 * H:].view(T, nh, hd).contiguous()
+            ctx = ops.attn_varlen(q, k, v, cu, bmap, scale=scale)
+            attn_out = ops.linear(ctx.reshape(T, H), layer.wo, layer.bo)
+            x = ops.layer_norm(attn_out, layer.ln1_w, layer.ln1_b,
+                               eps=d.layer_norm_eps, res=x)
+            act = ops.ACT_GELU_ERF if d.gelu == "erf" else ops.ACT_GELU_TANH
+            hmid = ops.linear(x, layer.w1, layer.b1, act=act)
+            out = ops.linear(hmid, layer.w2, layer.b2)
+            x = ops.layer_norm(out, layer.ln2_w, layer.ln2_b,
+                               eps=d.layer_norm_eps, res=x)
+        padded = torch.zeros(B, S, H, dtype=x.dtype, device=x.device)
+        padded[maskb] = x
+        return padded

@@ -477,3 +477,26 @@ def test_norm_small_batch_variant(device, rows):
                                      return_sum=True)
     _bf16_close(o2, ref2, atol=3e-2, rtol=3e-2)
     _bf16_close(s2, refs2, atol=3e-2, rtol=3e-2)
+
+
+def test_bert_packed_matches_padded(device):
+    """BERT's automatic varlen path vs the padded fp32 CPU forward."""
+    from distllm_amd.models import build_model
+
+    torch.manual_seed(61)
+    cpu = build_model("bert-tiny", device="cpu", dtype=torch.float32)
+    gpu = build_model("bert-tiny", device=device, dtype=torch.bfloat16)
+    gpu.load_state_dict(
+        {k: v.to(device, torch.bfloat16) for k, v in
+         cpu.state_dict().items()})
+    B, S = 4, 64
+    ids = torch.randint(4, 500, (B, S))
+    mask = torch.ones(B, S, dtype=torch.long)
+    for r, L in enumerate([64, 33, 9, 50]):
+        mask[r, L:] = 0
+        ids[r, L:] = 0
+    ref = cpu(ids, mask)
+    got = gpu(ids.to(device), mask.to(device))
+    m = mask.bool()
+    err = (got.float().cpu()[m] - ref.float()[m]).abs().max().item()
+    assert err < 0.08, f"packed bert vs padded fp32: max err {err}"
