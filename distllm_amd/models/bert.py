@@ -154,6 +154,13 @@ class BertEncoder(nn.Module):
         bmap = ops.varlen_block_map(cu)
         x = (self.word_emb[ids[maskb]] + self.pos_emb[pos.long()]
              + self.type_emb[0])
+        # pad the packed rows to a multiple of 256 so the projection
+        # GEMMs keep their autotuned fused-epilogue shapes; pad rows
+        # are inert (not in the block map; attn_varlen zeroes them)
+        T0 = x.shape[0]
+        Tp = (T0 + 255) // 256 * 256
+        if Tp != T0:
+            x = torch.nn.functional.pad(x, (0, 0, 0, Tp - T0))
         x = ops.layer_norm(x, self.emb_ln_w, self.emb_ln_b,
                            eps=d.layer_norm_eps)
         nh, hd = d.num_heads, d.head_dim
@@ -175,5 +182,5 @@ class BertEncoder(nn.Module):
             x = ops.layer_norm(out, layer.ln2_w, layer.ln2_b,
                                eps=d.layer_norm_eps, res=x)
         padded = torch.zeros(B, S, H, dtype=x.dtype, device=x.device)
-        padded[maskb] = x
+        padded[maskb] = x[:T0]
         return padded

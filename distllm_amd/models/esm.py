@@ -184,6 +184,11 @@ class Esm2Encoder(nn.Module, _RotaryMixin):
         cs, sn = self._rope(x.device)
         nh, hd = d.num_heads, d.head_dim
         H = d.hidden_size
+        T0 = xp.shape[0]
+        Tp = (T0 + 255) // 256 * 256
+        if Tp != T0:  # keep GEMMs on autotuned 256-aligned shapes
+            xp = torch.nn.functional.pad(xp, (0, 0, 0, Tp - T0))
+            pos = torch.nn.functional.pad(pos, (0, Tp - T0))
         T = xp.shape[0]
         for layer in self.layers:
             h = ops.layer_norm(xp, layer.ln1_w, layer.ln1_b,
@@ -204,7 +209,7 @@ class Esm2Encoder(nn.Module, _RotaryMixin):
         xp = ops.layer_norm(xp, self.final_ln_w, self.final_ln_b,
                             eps=d.layer_norm_eps)
         out = torch.zeros(B, S, H, dtype=xp.dtype, device=xp.device)
-        out[maskb] = xp
+        out[maskb] = xp[:T0]
         return out
 
 
@@ -307,6 +312,11 @@ class EsmCEncoder(nn.Module, _RotaryMixin):
         cs, sn = self._rope(x.device)
         nh, hd = d.num_heads, d.head_dim
         H = d.hidden_size
+        T0 = xp.shape[0]
+        Tp = (T0 + 255) // 256 * 256
+        if Tp != T0:
+            xp = torch.nn.functional.pad(xp, (0, 0, 0, Tp - T0))
+            pos = torch.nn.functional.pad(pos, (0, Tp - T0))
         T = xp.shape[0]
         rs = self.residue_scale
         for layer in self.layers:
@@ -329,5 +339,5 @@ class EsmCEncoder(nn.Module, _RotaryMixin):
         xp = ops.layer_norm(xp, self.final_ln_w, self.final_ln_b,
                             eps=d.layer_norm_eps)
         out = torch.zeros(B, S, H, dtype=xp.dtype, device=xp.device)
-        out[maskb] = xp
+        out[maskb] = xp[:T0]
         return out

@@ -296,7 +296,9 @@ torch::Tensor attn_varlen(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
   const int B = cu.size(0) - 1, nblocks = blk_seq.size(0);
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   TORCH_CHECK(H % HKV == 0, "GQA requires H % HKV == 0");
-  auto O = torch::empty_like(Q);
+  // zeros: rows past cu[B] (padding added to keep GEMM shapes 256-
+  // aligned) are not covered by the block map and must stay inert
+  auto O = torch::zeros_like(Q);
   launch_attn_varlen(Q.data_ptr(), K.data_ptr(), V.data_ptr(), O.data_ptr(),
                      cu.data_ptr<int>(), blk_seq.data_ptr<int>(),
                      blk_qb.data_ptr<int>(), nblocks, B, H, HKV, D,
