@@ -286,19 +286,7 @@ class LLMEngine:
             next_ids = self._sample_top_p(logits, params)
         elif len(seqs) <= 16 and params.temperature > 0 \
                 and str(dev).startswith("cuda"):
-            # small-batch Gumbel-max draw: the CDF-walk kernel is one
-            # block per row (latency-bound, ~170us at B=1 over a 128k
-            # vocab); argmax over logits/T + Gumbel noise is exact for
-            # categorical sampling and fully parallel. Seeded engine
-            # RNG keeps TP ranks identical.
-            lf = logits.float() / max(params.temperature, 1e-5)
-            if params.min_p > 0:
-                floor = (lf.max(dim=-1, keepdim=True).values
-                         + math.log(params.min_p))
-                lf = lf.masked_fill(lf < floor, float("-inf"))
-            u = torch.rand(lf.shape, device=dev, generator=self._rng)
-            g = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))
-            next_ids = (lf + g).argmax(dim=-1).int()
+            next_ids = self._gumbel_sample(logits, params, self._rng)
         else:
             rand = torch.rand(len(seqs), device=dev,
                               generator=self._rng)
@@ -328,6 +316,24 @@ class LLMEngine:
         for b in batch_sizes or [self.max_batch_size]:
             if self._graph_for_or_eager(b) is None:
                 return
+
+    @staticmethod
+    def _gumbel_sample(logits: torch.Tensor, params: SamplingParams,
+                       rng: torch.Generator) -> torch.Tensor:
+        """Small-batch Gumbel-max draw: the CDF-walk kernel is one
+        block per row (latency-bound, ~170us at B=1 over a 128k
+        vocab); argmax over logits/T + Gumbel noise is exact for
+        categorical sampling and fully parallel. Seeded engine RNG
+        keeps TP ranks identical."""
+        lf = logits.float() / max(params.temperature, 1e-5)
+        if params.min_p > 0:
+            floor = (lf.max(dim=-1, keepdim=True).values
+                     + math.log(params.min_p))
+            lf = lf.masked_fill(lf < floor, float("-inf"))
+        u = torch.rand(lf.shape, device=lf.device, generator=rng)
+        gumbel = -torch.log((-torch.log(u.clamp_min(1e-20)))
+                            .clamp_min(1e-20))
+        return (lf + gumbel).argmax(dim=-1).int()
 
     def _sample_top_p(self, logits: torch.Tensor,
                       params: SamplingParams) -> torch.Tensor:

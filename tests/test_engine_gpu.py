@@ -129,3 +129,18 @@ def test_small_batch_gumbel_sampling(model):
     out2 = engine2.generate([[5, 9, 100]], params)[0]
     assert out1 == out2  # same seeded RNG -> same draws
     assert len(out1) == 6
+
+    # the Gumbel noise must actually randomize: two equal-probability
+    # spikes must BOTH be drawn across many samples (catches the
+    # constant-noise regression where sampling collapses to greedy)
+    import torch
+
+    rng = torch.Generator(device="cuda").manual_seed(7)
+    logits = torch.full((64, 32), -20.0, device="cuda")
+    logits[:, 5] = 2.0
+    logits[:, 9] = 2.0
+    p2 = SamplingParams(temperature=1.0, min_p=0.0, top_p=1.0)
+    got = set()
+    for _ in range(4):
+        got |= set(LLMEngine._gumbel_sample(logits, p2, rng).tolist())
+    assert got == {5, 9}, f"gumbel draws collapsed: {got}"
