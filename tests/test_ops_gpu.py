@@ -354,11 +354,12 @@ def test_attn_varlen_d128_gqa(device):
     _bf16_close(out, ref, atol=3e-2, rtol=3e-2)
 
 
-def test_esm2_packed_matches_padded(device):
+def test_esm2_packed_matches_padded(device, monkeypatch):
     """The encoder's automatic varlen path must agree with the padded
     CPU fp32 forward on a mixed-length protein batch."""
     from distllm_amd.models import build_model
 
+    monkeypatch.setenv("DISTLLM_FORCE_VARLEN", "1")
     torch.manual_seed(33)
     cpu = build_model("esm2-8m", device="cpu", dtype=torch.float32)
     gpu = build_model("esm2-8m", device=device, dtype=torch.bfloat16)
@@ -479,10 +480,11 @@ def test_norm_small_batch_variant(device, rows):
     _bf16_close(s2, refs2, atol=3e-2, rtol=3e-2)
 
 
-def test_bert_packed_matches_padded(device):
+def test_bert_packed_matches_padded(device, monkeypatch):
     """BERT's automatic varlen path vs the padded fp32 CPU forward."""
     from distllm_amd.models import build_model
 
+    monkeypatch.setenv("DISTLLM_FORCE_VARLEN", "1")
     torch.manual_seed(61)
     cpu = build_model("bert-tiny", device="cpu", dtype=torch.float32)
     gpu = build_model("bert-tiny", device=device, dtype=torch.bfloat16)
