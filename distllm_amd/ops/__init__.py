@@ -256,7 +256,11 @@ def linear(x, weight, bias=None, act=ACT_NONE):
             elif mode == "gemm16":
                 out = _hip_ops.gemm16_bf16(x2, weight, bias, act)
             else:  # auto
-                key = (min(M, 1 << 20) // 4096, N, K, act)
+                # coarse bucket above 16k rows: packed varlen batches
+                # produce a different M per batch, and a fresh
+                # 4-implementation autotune probe per distinct M
+                # (~100ms each) dominated the semantic-chunk pipeline
+                key = ("big" if M >= 16384 else M // 4096, N, K, act)
                 choice = _GEMM_CHOICE.get(key)
                 if choice is None:
                     choice = _autotune_gemm(x2, weight, bias, act)
