@@ -114,19 +114,16 @@ class BertEncoder(nn.Module):
 
         ids = input_ids.long()
         B, S = ids.shape
+        # BERT packed/varlen is opt-in only: on its realistic padded
+        # workloads (semantic-chunk buffers AND chunk re-embeds, 12
+        # light layers, ~200-400-token rows) the padded path measured
+        # FASTER (73 vs 47-50 docs/s on the pipeline bench) — the
+        # per-layer pack overheads beat the skipped pad FLOPs at this
+        # depth/width. The deep/wide ESM encoders auto-pack instead
+        # (x3.0 measured, profiles/RESULTS.md).
         if (ids.is_cuda and ops.HAVE_HIP and attention_mask is not None
-                and not os.environ.get("DISTLLM_NO_VARLEN")
                 and token_type_ids is None
-                and (os.environ.get("DISTLLM_FORCE_VARLEN")
-                     or (float(attention_mask.float().mean()) < 0.7
-                         and float(attention_mask.float().sum())
-                         >= 256 * B))):
-            # packed pays when padding is substantial AND sequences are
-            # long enough to amortize the per-layer pack overheads:
-            # short-buffer semantic-chunk batches (mean ~190 tokens)
-            # measured 73 docs/s padded vs 46.6 packed, while the
-            # length-dispersed protein batches (mean 344 of 1024)
-            # measured 2.66x FASTER packed (profiles/RESULTS.md)
+                and os.environ.get("DISTLLM_FORCE_VARLEN")):
             return self._forward_packed(ids, attention_mask)
         pos = torch.arange(S, device=ids.device)
         x = self.word_emb[ids] + self.pos_emb[pos][None, :, :]
