@@ -28,7 +28,7 @@ os.environ.setdefault("DISTLLM_ALLOW_HASH_TOKENIZER", "1")
 def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch-size", type=int, default=512)
     ap.add_argument("--seq-len", type=int, default=512)
