@@ -114,8 +114,12 @@ def generate(
         "max_tokens": max_tokens,
         "max_batch_size": batch_size,
     }
-    if generator_name in ("openai_http", "argo", "langchain"):
+    if generator_name in ("openai_http", "argo"):
         generator_kwargs = {"name": generator_name, "model": llm_name,
+                            "temperature": temperature,
+                            "max_tokens": max_tokens}
+    elif generator_name in ("langchain", "api"):
+        generator_kwargs = {"name": generator_name, "llm": llm_name,
                             "temperature": temperature,
                             "max_tokens": max_tokens}
     output_dir.mkdir(parents=True, exist_ok=True)

@@ -166,7 +166,7 @@ class OpenAICompatGeneratorConfig(BaseConfig):
     """HTTP client against any OpenAI-compatible /v1 server (our
     chat_server, a remote endpoint, or an Argo-style proxy)."""
 
-    name: Literal["openai_http", "argo", "langchain"] = "openai_http"
+    name: Literal["openai_http", "argo"] = "openai_http"
     base_url: str = "http://localhost:8000/v1"
     model: str = "default"
     api_key_env: str = Field(
@@ -222,13 +222,131 @@ class OpenAICompatGenerator:
 
 GeneratorConfigs = Union[NativeGeneratorConfig, OpenAICompatGeneratorConfig]
 
+
+
+class ApiGeneratorConfig(BaseConfig):
+    """API-model generation (reference `langchain` backend parity:
+    distllm/generate/generators/langchain_backend.py — gpt-3.5-turbo /
+    gemini-pro / claude-3-opus via API keys in a .env file). SDK-free:
+    the three wire protocols are spoken directly over httpx."""
+
+    name: Literal["langchain", "api"] = "langchain"
+    llm: str = Field("gpt-3.5-turbo",
+                     description="gpt-* | gemini-* | claude-* model id")
+    temperature: float = 0.0
+    max_tokens: int = 2000
+    verbose: bool = True
+    dotenv_path: str = "~/.env"
+    timeout: float = 120.0
+
+
+class ApiGenerator:
+    """Dispatches on the model-id prefix to the OpenAI, Google
+    Generative Language or Anthropic Messages REST APIs."""
+
+    def __init__(self, config: ApiGeneratorConfig):
+        self.config = config
+        self._load_dotenv(config.dotenv_path)
+
+    @staticmethod
+    def _load_dotenv(path: str) -> None:
+        """Minimal .env loader (python-dotenv-free): KEY=VALUE lines,
+        existing environment wins (reference load_dotenv semantics)."""
+        import pathlib
+
+        p = pathlib.Path(path).expanduser()
+        if not p.exists():
+            return
+        for line in p.read_text().splitlines():
+            line = line.strip()
+            if not line or line.startswith("#") or "=" not in line:
+                continue
+            k, _, v = line.partition("=")
+            os.environ.setdefault(k.strip(), v.strip().strip("'\""))
+
+    def build_request(self, prompt: str):
+        """(url, headers, json_body, parse) for one prompt — separated
+        from the HTTP send so the wire formats are unit-testable
+        offline."""
+        c = self.config
+        m = c.llm
+        if m.startswith("gemini"):
+            key = os.environ.get("GOOGLE_API_KEY", "")
+            url = ("https://generativelanguage.googleapis.com/v1beta/"
+                   f"models/{m}:generateContent?key={key}")
+            body = {
+                "contents": [{"parts": [{"text": prompt}]}],
+                "generationConfig": {"temperature": c.temperature,
+                                     "maxOutputTokens": c.max_tokens},
+            }
+
+            def parse(j):
+                return j["candidates"][0]["content"]["parts"][0]["text"]
+
+            return url, {"Content-Type": "application/json"}, body, parse
+        if m.startswith("claude"):
+            url = "https://api.anthropic.com/v1/messages"
+            headers = {
+                "x-api-key": os.environ.get("ANTHROPIC_API_KEY", ""),
+                "anthropic-version": "2023-06-01",
+                "Content-Type": "application/json",
+            }
+            body = {
+                "model": m,
+                "max_tokens": c.max_tokens,
+                "temperature": c.temperature,
+                "messages": [{"role": "user", "content": prompt}],
+            }
+
+            def parse(j):
+                return j["content"][0]["text"]
+
+            return url, headers, body, parse
+        # default: OpenAI chat completions (gpt-*)
+        url = "https://api.openai.com/v1/chat/completions"
+        headers = {
+            "Authorization":
+                f"Bearer {os.environ.get('OPENAI_API_KEY', '')}",
+            "Content-Type": "application/json",
+        }
+        body = {
+            "model": m,
+            "temperature": c.temperature,
+            "max_tokens": c.max_tokens,
+            "messages": [{"role": "user", "content": prompt}],
+        }
+
+        def parse(j):
+            return j["choices"][0]["message"]["content"]
+
+        return url, headers, body, parse
+
+    @retry(tries=3, base_delay=1.0)
+    def _send(self, url, headers, body, parse) -> str:
+        import httpx
+
+        r = httpx.post(url, headers=headers, json=body,
+                       timeout=self.config.timeout)
+        r.raise_for_status()
+        return parse(r.json())
+
+    def generate(self, prompts):
+        if isinstance(prompts, str):
+            prompts = [prompts]
+        out = []
+        for p in prompts:
+            out.append(self._send(*self.build_request(p)))
+        return out
+
+
 STRATEGIES: dict[str, tuple[type, type]] = {
     "native": (NativeGeneratorConfig, NativeGenerator),
     "vllm": (NativeGeneratorConfig, NativeGenerator),
     "huggingface": (NativeGeneratorConfig, NativeGenerator),
     "openai_http": (OpenAICompatGeneratorConfig, OpenAICompatGenerator),
     "argo": (OpenAICompatGeneratorConfig, OpenAICompatGenerator),
-    "langchain": (OpenAICompatGeneratorConfig, OpenAICompatGenerator),
+    "langchain": (ApiGeneratorConfig, ApiGenerator),
+    "api": (ApiGeneratorConfig, ApiGenerator),
 }
 
 

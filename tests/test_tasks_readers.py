@@ -104,3 +104,41 @@ def test_huggingface_reader_roundtrip(tmp_path):
     rd = get_reader({"name": "huggingface"})
     text, paths = rd.read(tmp_path / "ds")
     assert text == ["a", "b"] and paths == ["p0", "p1"]
+
+
+def test_api_generator_wire_formats(tmp_path, monkeypatch):
+    """The `langchain`-parity API generator speaks the three wire
+    protocols (OpenAI / Google Generative Language / Anthropic
+    Messages) without SDKs; request construction is checked offline."""
+    from distllm_amd.generate import get_generator
+
+    envf = tmp_path / "env"
+    envf.write_text("OPENAI_API_KEY=okey\nGOOGLE_API_KEY=gkey\n"
+                    "ANTHROPIC_API_KEY=akey\n")
+    monkeypatch.delenv("OPENAI_API_KEY", raising=False)
+    monkeypatch.delenv("GOOGLE_API_KEY", raising=False)
+    monkeypatch.delenv("ANTHROPIC_API_KEY", raising=False)
+
+    g = get_generator({"name": "langchain", "llm": "gpt-3.5-turbo",
+                       "dotenv_path": str(envf), "temperature": 0.0})
+    url, hdr, body, parse = g.build_request("hi")
+    assert url.endswith("/v1/chat/completions")
+    assert hdr["Authorization"] == "Bearer okey"
+    assert body["messages"] == [{"role": "user", "content": "hi"}]
+    assert parse({"choices": [{"message": {"content": "x"}}]}) == "x"
+
+    g2 = get_generator({"name": "langchain", "llm": "gemini-pro",
+                        "dotenv_path": str(envf)})
+    url, hdr, body, parse = g2.build_request("q")
+    assert "generativelanguage.googleapis.com" in url and "key=gkey" in url
+    assert body["contents"][0]["parts"][0]["text"] == "q"
+    assert parse({"candidates": [{"content": {"parts": [{"text": "y"}]}}]}) == "y"
+
+    g3 = get_generator({"name": "langchain",
+                        "llm": "claude-3-opus-20240229",
+                        "dotenv_path": str(envf)})
+    url, hdr, body, parse = g3.build_request("p")
+    assert url == "https://api.anthropic.com/v1/messages"
+    assert hdr["x-api-key"] == "akey"
+    assert hdr["anthropic-version"] == "2023-06-01"
+    assert parse({"content": [{"text": "z"}]}) == "z"
