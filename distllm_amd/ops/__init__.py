@@ -210,6 +210,10 @@ def linear(x, weight, bias=None, act=ACT_NONE):
         mode = os.environ.get("DISTLLM_GEMM", "auto")
         if (mode != "blaslt" and act == ACT_NONE and M <= 256
                 and N % 64 == 0 and K % 64 == 0 and x2.is_contiguous()):
+            if mode == "decode":  # forced in-chain A/B (profiling)
+                sk = _decode_splitk(M, N, K)
+                out = _hip_ops.decode_gemm(x2, weight, bias, sk)
+                return out.reshape(*orig_shape[:-1], N)
             # bucket by padded M so varying decode/prefill
             # batch sizes reuse one tuning decision
             key = ("dec", 128 if M <= 128 else 256, N, K,
