@@ -48,7 +48,7 @@ def resolve_preset(name_or_path: str) -> str:
     if "mistral" in low:
         return "mistral-7b"
     if "llama" in low:
-        return "llama3-8b"
+        return "llama3-70b" if "70" in low else "llama3-8b"
     raise ValueError(f"cannot resolve model preset for {name_or_path!r}")
 
 

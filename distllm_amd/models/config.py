@@ -94,6 +94,14 @@ PRESETS: dict[str, ModelDims] = {
         num_heads=32, num_kv_heads=8, intermediate_size=14336,
         max_position_embeddings=8192, rms_norm_eps=1e-5, rope_theta=500000.0,
     ),
+    # 140 GB of bf16 weights: fits — with KV headroom — on ONE 288 GB
+    # MI355X, no tensor parallelism required (the reference's vLLM
+    # configs needed TP8 for far smaller models on 80 GB cards)
+    "llama3-70b": ModelDims(
+        arch="llama", vocab_size=128256, hidden_size=8192, num_layers=80,
+        num_heads=64, num_kv_heads=8, intermediate_size=28672,
+        max_position_embeddings=8192, rms_norm_eps=1e-5, rope_theta=500000.0,
+    ),
     "llama-tiny": ModelDims(
         arch="llama", vocab_size=2048, hidden_size=256, num_layers=2,
         num_heads=4, num_kv_heads=2, intermediate_size=512,
