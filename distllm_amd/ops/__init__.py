@@ -195,10 +195,10 @@ def _decode_splitk(M, N, K):
 
 def linear(x, weight, bias=None, act=ACT_NONE):
     """Linear layer entry used by models. GPU path: autotuned choice
-    between hipBLASLt (+fused activation kernel), the hand-written
-    deep-pipelined MFMA GEMM (gemm8.hip), and the skinny-M split-K
-    decode GEMM (decode_gemm.hip) for M <= 256. Override with
-    DISTLLM_GEMM=blaslt|gemm8."""
+    between hipBLASLt (plain and fused-epilogue), the hand-written
+    MFMA GEMMs (gemm8.hip, 8-phase gemm16.hip) and the skinny-M
+    split-K decode GEMM (decode_gemm.hip) for M <= 256. Override with
+    DISTLLM_GEMM=blaslt|gemm8|gemm16|decode."""
     import os
 
     orig_shape = x.shape
