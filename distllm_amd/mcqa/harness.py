@@ -165,12 +165,23 @@ class LocalServer:
         raise TimeoutError("local server failed to become healthy")
 
     def stop(self) -> None:
+        """Graceful-first shutdown: SIGINT lets uvicorn close and the
+        process destroy its GPU context cleanly — hard-killing a
+        process mid-kernel can leave the device queues in a state that
+        intermittently aborts LATER GPU work in this process (observed
+        as full-test-suite flakes surfacing tests later)."""
         if self.proc is not None and self.proc.poll() is None:
-            self.proc.terminate()
+            import signal
+
+            self.proc.send_signal(signal.SIGINT)
             try:
-                self.proc.wait(timeout=10)
+                self.proc.wait(timeout=15)
             except subprocess.TimeoutExpired:
-                self.proc.kill()
+                self.proc.terminate()
+                try:
+                    self.proc.wait(timeout=10)
+                except subprocess.TimeoutExpired:
+                    self.proc.kill()
         self.proc = None
 
 
