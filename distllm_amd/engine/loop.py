@@ -40,6 +40,17 @@ class EngineLoop:
         self._waiting: list[_Request] = []
         self._running: list[_Request] = []
         self._closed = False
+        # pre-capture every pow2 decode-graph bucket BEFORE serving:
+        # capturing graphs mid-flight (batch bucket shrinks as requests
+        # drain) while submitter threads run was observed to
+        # intermittently abort the HIP runtime under stress
+        if engine.use_graphs:
+            buckets, b = [], 1
+            while b < engine.max_batch_size:
+                buckets.append(b)
+                b <<= 1
+            buckets.append(engine.max_batch_size)
+            engine.warm_graphs(sorted(set(buckets)))
         self._thread = threading.Thread(target=self._loop, daemon=True)
         self._thread.start()
 
