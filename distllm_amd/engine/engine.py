@@ -362,6 +362,19 @@ class _GraphRunner:
     overhead that dominates small-batch decode.
     """
 
+    def __del__(self):
+        # Drain the device before torch destroys the captured hipGraph:
+        # GC can collect a dead engine's graphs from any thread at any
+        # time, and hipGraphExecDestroy racing in-flight kernels was
+        # the best candidate for intermittent full-test-suite aborts.
+        try:
+            import torch as _t
+
+            if _t.cuda.is_available():
+                _t.cuda.synchronize()
+        except Exception:
+            pass
+
     def __init__(self, engine: LLMEngine, batch: int):
         dev = engine.device
         self.engine = engine
