@@ -13,6 +13,7 @@ Inference-only, bf16 weights, fp32 accumulation inside kernels.
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 from torch import nn
@@ -110,7 +111,6 @@ class BertEncoder(nn.Module):
         attention_mask: torch.Tensor,
         token_type_ids: torch.Tensor | None = None,
     ) -> torch.Tensor:
-        import os
 
         ids = input_ids.long()
         B, S = ids.shape

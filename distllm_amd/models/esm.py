@@ -16,6 +16,7 @@ bf16 weights, fp32 accumulation in kernels.
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 from torch import nn
@@ -103,7 +104,6 @@ class Esm2Encoder(nn.Module, _RotaryMixin):
         ids = input_ids.long()
         B, S = ids.shape
         x = self.word_emb[ids]
-        import os
         if (x.is_cuda and ops.HAVE_HIP and attention_mask is not None
                 and not os.environ.get("DISTLLM_NO_VARLEN")
                 and (os.environ.get("DISTLLM_FORCE_VARLEN")
@@ -266,7 +266,6 @@ class EsmCEncoder(nn.Module, _RotaryMixin):
         ids = input_ids.long()
         B, S = ids.shape
         x = self.word_emb[ids]
-        import os
         if (x.is_cuda and ops.HAVE_HIP and attention_mask is not None
                 and not os.environ.get("DISTLLM_NO_VARLEN")
                 and (os.environ.get("DISTLLM_FORCE_VARLEN")
