@@ -11,6 +11,8 @@ weights (no network), answer length capped like MCQA answers.
   python profiles/rag_e2e_bench.py --questions 64 --corpus 100000
 """
 
+import os
+os.environ.setdefault("DISTLLM_ALLOW_HASH_TOKENIZER", "1")
 import argparse
 import sys
 import time
