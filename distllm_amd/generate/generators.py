@@ -100,6 +100,16 @@ class NativeGenerator:
                                 max_batch_size=config.max_batch_size,
                                 max_model_len=config.max_model_len,
                                 kv_cache_dtype=config.kv_cache_dtype)
+        # vLLM-style capture-at-init: pre-record every pow2 decode
+        # graph bucket so batch drain never captures inside a timed /
+        # serving window (measured ~0.2s per first-use capture)
+        if self.engine.use_graphs:
+            b, buckets = 1, []
+            while b < config.max_batch_size:
+                buckets.append(b)
+                b <<= 1
+            buckets.append(config.max_batch_size)
+            self.engine.warm_graphs(sorted(set(buckets)))
         self._loop = None
         eos = getattr(self.tokenizer, "eos_token_id", None)
         self.params = SamplingParams(
