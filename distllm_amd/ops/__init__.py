@@ -342,14 +342,18 @@ def varlen_block_map(cu_seqlens: torch.Tensor, qblk: int = 128):
     cu = cu_seqlens.to("cpu", torch.int64)
     lens = cu[1:] - cu[:-1]
     nqb = torch.div(lens + qblk - 1, qblk, rounding_mode="floor")
+    total = int(nqb.sum())
     blk_seq = torch.repeat_interleave(
-        torch.arange(len(lens), dtype=torch.int32), nqb
+        torch.arange(len(lens), dtype=torch.int64), nqb
     )
-    blk_qb = torch.cat(
-        [torch.arange(int(n), dtype=torch.int32) for n in nqb]
-    ) if len(lens) else torch.zeros(0, dtype=torch.int32)
+    # vectorized within-sequence block index: position minus the
+    # sequence's first position
+    starts = torch.zeros(len(lens), dtype=torch.int64)
+    if len(lens) > 1:
+        starts[1:] = nqb.cumsum(0)[:-1]
+    blk_qb = torch.arange(total, dtype=torch.int64) -         torch.repeat_interleave(starts, nqb)
     dev = cu_seqlens.device
-    return blk_seq.to(dev), blk_qb.to(dev)
+    return blk_seq.int().to(dev), blk_qb.int().to(dev)
 
 
 def attn_varlen(Q, K, V, cu_seqlens, block_map=None, scale=None):

@@ -29,9 +29,14 @@ def bench(model, ids, mask, iters=8):
 
 
 def main():
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="esm2-650m")
+    a = ap.parse_args()
     assert torch.cuda.is_available() and ops.HAVE_HIP
     torch.manual_seed(0)
-    model = build_model("esm2-650m", device="cuda", dtype=torch.bfloat16)
+    model = build_model(a.model, device="cuda", dtype=torch.bfloat16)
     B, S = 128, 1024
     g = torch.Generator().manual_seed(7)
     # dispersed protein lengths: lognormal-ish, 30..1024
