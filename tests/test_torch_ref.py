@@ -250,3 +250,21 @@ def test_attn_varlen_ref_matches_padded():
     bs2, bq2 = varlen_block_map(
         torch.tensor([0, 130, 386], dtype=torch.int32))
     assert bs2.tolist() == [0, 0, 1, 1] and bq2.tolist() == [0, 1, 0, 1]
+
+
+def test_varlen_block_map_edge_cases():
+    """Zero sequences, zero-length sequences, and exact block
+    multiples all yield a consistent (seq, qblock) map."""
+    from distllm_amd.ops import varlen_block_map
+
+    def m(cu):
+        bs, bq = varlen_block_map(
+            torch.tensor(cu, dtype=torch.int32), 64)
+        return bs.tolist(), bq.tolist()
+
+    assert m([0]) == ([], [])
+    assert m([0, 1]) == ([0], [0])
+    assert m([0, 64]) == ([0], [0])          # exact multiple: 1 block
+    assert m([0, 65]) == ([0, 0], [0, 1])
+    # zero-length sequence in the middle is skipped entirely
+    assert m([0, 64, 64, 200]) == ([0, 2, 2, 2], [0, 0, 1, 2])
