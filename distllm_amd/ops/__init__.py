@@ -328,7 +328,7 @@ def attn_fwd(Q, K, V, mask=None, causal=False, scale=None):
 
 
 def ivf_score(slabs, lens, probes, q):
-    """Batched IVF chunk scoring: slabs [nchunks, cap, H] bf16, lens
+    """Batched IVF chunk scoring: slabs [nchunks, cap, H] f32, lens
     [nchunks] i32, probes [Q, nprobe] i32, q [Q, H] f32 ->
     [Q, nprobe, cap] f32 (-inf at pad rows)."""
     _require_hip()

@@ -111,3 +111,59 @@ def test_fuzz_paged_attention(device):
         out = ops.paged_attn(q, kc, vc, bt, ctx)
         ref = torch_ref.paged_attn(q, kc, vc, bt, ctx, 1 / math.sqrt(D))
         _close(out, ref)
+
+
+def test_fuzz_varlen_attention(device):
+    """Random dispersed length mixes (incl. zero-length and >128-block
+    sequences) vs the per-sequence fp32 reference."""
+    g = torch.Generator().manual_seed(321)
+    for it in range(10):
+        B = int(torch.randint(1, 7, (1,), generator=g))
+        lens = [int(torch.randint(0, 400, (1,), generator=g))
+                for _ in range(B)]
+        if sum(lens) == 0:
+            lens[0] = 5
+        H = [2, 4, 8][int(torch.randint(0, 3, (1,), generator=g))]
+        D = [64, 128][int(torch.randint(0, 2, (1,), generator=g))]
+        gqa = int(torch.randint(0, 2, (1,), generator=g))
+        HKV = H // 2 if gqa else H
+        cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                          dtype=torch.int32, device=device)
+        T = sum(lens)
+        q = torch.randn(T, H, D, device=device, dtype=torch.bfloat16)
+        k = torch.randn(T, HKV, D, device=device, dtype=torch.bfloat16)
+        v = torch.randn_like(k)
+        out = ops.attn_varlen(q, k, v, cu)
+        rep = H // HKV
+        ref = torch_ref.attn_varlen(
+            q, k.repeat_interleave(rep, dim=1),
+            v.repeat_interleave(rep, dim=1), cu, 1.0 / math.sqrt(D))
+        _close(out, ref)
+
+
+def test_fuzz_ivf_score(device):
+    """Random chunk counts/caps/dims (x4-aligned) and probe lists with
+    duplicates vs the einsum reference."""
+    g = torch.Generator().manual_seed(654)
+    for _ in range(10):
+        nch = int(torch.randint(1, 12, (1,), generator=g))
+        cap = int(torch.randint(1, 140, (1,), generator=g))
+        H = 4 * int(torch.randint(1, 100, (1,), generator=g))
+        Q = int(torch.randint(1, 9, (1,), generator=g))
+        nprobe = int(torch.randint(1, min(nch, 4) + 1, (1,), generator=g))
+        lens = torch.randint(0, cap + 1, (nch,), generator=g).int()
+        slabs = torch.randn(nch, cap, H, generator=g)
+        valid = torch.arange(cap)[None, :] < lens[:, None]
+        slabs = (slabs * valid.unsqueeze(-1)).to(device)
+        lens_d = lens.to(device)
+        probes = torch.randint(0, nch, (Q, nprobe), generator=g) \
+            .int().to(device)
+        qv = torch.randn(Q, H, generator=g).to(device)
+        got = ops.ivf_score(slabs, lens_d, probes, qv)
+        ref = torch.einsum("qpcd,qd->qpc", slabs[probes.long()], qv)
+        ref = ref.masked_fill(
+            ~(torch.arange(cap, device=device)[None, None, :]
+              < lens_d[probes.long()][:, :, None]), float("-inf"))
+        fin = torch.isfinite(ref)
+        assert torch.equal(torch.isfinite(got), fin)
+        assert torch.allclose(got[fin], ref[fin], atol=2e-3, rtol=2e-3)
