@@ -49,6 +49,8 @@ def resolve_preset(name_or_path: str) -> str:
         return "mistral-7b"
     if "llama" in low:
         return "llama3-70b" if "70" in low else "llama3-8b"
+    if "qwen" in low:
+        return "qwen2-7b"
     raise ValueError(f"cannot resolve model preset for {name_or_path!r}")
 
 

@@ -34,7 +34,7 @@ def build_model(name_or_dims: str | ModelDims, device: str | None = None,
         m = Esm2Encoder(dims, dtype=dtype, **kw)
     elif dims.arch == "esmc":
         m = EsmCEncoder(dims, dtype=dtype, **kw)
-    elif dims.arch in ("mistral", "llama"):
+    elif dims.arch in ("mistral", "llama", "qwen2"):
         m = DecoderModel(dims, dtype=dtype, **kw)
     else:
         raise ValueError(f"unknown arch {dims.arch}")

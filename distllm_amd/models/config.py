@@ -19,7 +19,7 @@ from pydantic import BaseModel
 
 
 class ModelDims(BaseModel):
-    arch: Literal["bert", "esm2", "esmc", "mistral", "llama"]
+    arch: Literal["bert", "esm2", "esmc", "mistral", "llama", "qwen2"]
     vocab_size: int
     hidden_size: int
     num_layers: int
@@ -33,6 +33,7 @@ class ModelDims(BaseModel):
     type_vocab_size: int = 2          # BERT token types
     gelu: Literal["tanh", "erf"] = "erf"
     tie_word_embeddings: bool = False
+    attention_bias: bool = False      # Qwen2: bias on q/k/v only
 
     @property
     def head_dim(self) -> int:
@@ -106,6 +107,19 @@ PRESETS: dict[str, ModelDims] = {
         arch="llama", vocab_size=2048, hidden_size=256, num_layers=2,
         num_heads=4, num_kv_heads=2, intermediate_size=512,
         max_position_embeddings=512,
+    ),
+    # Qwen2-7B-Instruct: llama-family + qkv bias (HF Qwen2ForCausalLM)
+    "qwen2-7b": ModelDims(
+        arch="qwen2", vocab_size=152064, hidden_size=3584, num_layers=28,
+        num_heads=28, num_kv_heads=4, intermediate_size=18944,
+        max_position_embeddings=32768, rms_norm_eps=1e-6,
+        rope_theta=1_000_000.0, attention_bias=True,
+    ),
+    "qwen2-tiny": ModelDims(
+        arch="qwen2", vocab_size=1024, hidden_size=256, num_layers=2,
+        num_heads=4, num_kv_heads=2, intermediate_size=512,
+        max_position_embeddings=512, rms_norm_eps=1e-6,
+        attention_bias=True,
     ),
 }
 

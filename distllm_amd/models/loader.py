@@ -92,6 +92,12 @@ def _map_llama(src: dict, model) -> dict:
             rows(src[lp + "self_attn.k_proj.weight"], d.kv_heads * hd),
             rows(src[lp + "self_attn.v_proj.weight"], d.kv_heads * hd),
         ], 0)
+        if lp + "self_attn.q_proj.bias" in src:  # Qwen2-family
+            out[f"layers.{i}.bqkv"] = torch.cat([
+                rows(src[lp + "self_attn.q_proj.bias"], d.num_heads * hd),
+                rows(src[lp + "self_attn.k_proj.bias"], d.kv_heads * hd),
+                rows(src[lp + "self_attn.v_proj.bias"], d.kv_heads * hd),
+            ], 0)
         out[f"layers.{i}.wo"] = cols(src[lp + "self_attn.o_proj.weight"],
                                      d.num_heads * hd)
         out[f"layers.{i}.in_norm"] = src[lp + "input_layernorm.weight"]
@@ -158,7 +164,7 @@ def load_weights(model, path: str | Path) -> None:
         arch = model.dims.arch
         if arch == "bert":
             mapped = _map_bert(src, model)
-        elif arch in ("llama", "mistral"):
+        elif arch in ("llama", "mistral", "qwen2"):
             mapped = _map_llama(src, model)
         elif arch == "esm2":
             mapped = _map_esm2(src, model)

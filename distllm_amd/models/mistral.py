@@ -42,6 +42,9 @@ class DecoderLayer(nn.Module):
         self._nq, self._nkv = nh * hd, kvh * hd
         self._inter = inter
         self.wqkv = nn.Parameter(torch.empty((nh + 2 * kvh) * hd, H))
+        # Qwen2-family: bias on the q/k/v projections only
+        self.bqkv = (nn.Parameter(torch.empty((nh + 2 * kvh) * hd))
+                     if d.attention_bias else None)
         self.wo = nn.Parameter(torch.empty(H, nh * hd))
         self.in_norm = nn.Parameter(torch.empty(H))
         self.post_norm = nn.Parameter(torch.empty(H))
@@ -97,7 +100,7 @@ class DecoderModel(nn.Module):
         )
         self.final_norm = nn.Parameter(torch.empty(H))
         if with_lm_head is None:
-            with_lm_head = dims.arch == "llama"
+            with_lm_head = dims.arch in ("llama", "qwen2")
         self.lm_head = (
             nn.Parameter(torch.empty(dims.vocab_size, H))
             if with_lm_head and not dims.tie_word_embeddings
@@ -113,6 +116,8 @@ class DecoderModel(nn.Module):
         for n, p in self.named_parameters():
             if p.dim() >= 2:
                 nn.init.normal_(p, std=0.02)
+            elif n.endswith("bqkv"):
+                nn.init.zeros_(p)
             else:
                 nn.init.ones_(p)
 
@@ -163,7 +168,8 @@ class DecoderModel(nn.Module):
         nh, kvh, hd = self.local_heads, self.local_kv_heads, d.head_dim
         cs, sn = self._rope(h.device)
         # ONE fused qkv GEMM, ONE rope launch (q|k roped, v copied)
-        qkv = ops.linear(h, layer.wqkv).view(B, S, nh + 2 * kvh, hd)
+        qkv = ops.linear(h, layer.wqkv,
+                         layer.bqkv).view(B, S, nh + 2 * kvh, hd)
         qkv = ops.rope_apply(qkv, cs, sn, pos, rope_heads=nh + kvh)
         q = qkv[:, :, :nh]
         k = qkv[:, :, nh: nh + kvh]
@@ -242,7 +248,8 @@ class DecoderModel(nn.Module):
         for li, layer in enumerate(self.layers):
             # ONE fused qkv GEMM + ONE rope (q|k roped, v copied):
             # 5 launches -> 2 on the per-token critical path
-            qkv = ops.linear(h, layer.wqkv).view(B, 1, nh + 2 * kvh, hd)
+            qkv = ops.linear(h, layer.wqkv,
+                             layer.bqkv).view(B, 1, nh + 2 * kvh, hd)
             qkv = ops.rope_apply(qkv, cs, sn, pos, rope_heads=nh + kvh)
             q = qkv[:, 0, :nh]
             k = qkv[:, 0, nh: nh + kvh].contiguous()

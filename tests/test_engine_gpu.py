@@ -144,3 +144,26 @@ def test_small_batch_gumbel_sampling(model):
     for _ in range(4):
         got |= set(LLMEngine._gumbel_sample(logits, p2, rng).tolist())
     assert got == {5, 9}, f"gumbel draws collapsed: {got}"
+
+
+def test_qwen2_engine_greedy_gpu(device):
+    """Qwen2-family (qkv bias) through the full paged-decode engine:
+    greedy continuation must match the teacher-forced full forward up
+    to bf16 near-ties."""
+    torch.manual_seed(8)
+    m = build_model("qwen2-tiny", device="cuda:0", dtype=torch.bfloat16)
+    with torch.no_grad():  # random bias, not the zero init
+        for lyr in m.layers:
+            lyr.bqkv.normal_(std=0.1)
+    engine = LLMEngine(m, max_batch_size=4, max_model_len=128,
+                       num_blocks=128)
+    params = SamplingParams(temperature=0.0, min_p=0.0, max_tokens=8)
+    prompts = [[5, 9, 100, 42], list(range(10, 30))]
+    outs = engine.generate(prompts, params)
+    bad = []
+    for p, o in zip(prompts, outs):
+        ref, margins = _ref_greedy(m, p, len(o), follow=o)
+        for a, b, mg in zip(o, ref, margins):
+            if a != b and mg > 0.05:
+                bad.append((p[:3], a, b, mg))
+    assert not bad, f"clear-margin disagreements: {bad}"

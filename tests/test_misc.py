@@ -235,3 +235,26 @@ def test_example_yamls_parse(tmp_path):
     # evaluate.yaml is a plain task-list document for `distllm evaluate`
     ev = yaml.safe_load((root / "evaluate.yaml").read_text())
     assert "tasks" in ev or "task_names" in ev or isinstance(ev, dict)
+
+
+def test_qwen2_preset_and_cpu_forward():
+    """Qwen2 family resolves, builds, and runs the fp32 CPU path with
+    the fused qkv bias applied."""
+    import torch
+
+    from distllm_amd.embed.encoders import resolve_preset
+    from distllm_amd.models import build_model
+
+    assert resolve_preset("Qwen/Qwen2-7B-Instruct") == "qwen2-7b"
+    torch.manual_seed(0)
+    m = build_model("qwen2-tiny", device="cpu", dtype=torch.float32)
+    assert m.lm_head is not None  # causal-LM head by default
+    ids = torch.randint(4, 1024, (2, 12))
+    mask = torch.ones(2, 12, dtype=torch.long)
+    h = m(ids, mask)
+    assert h.shape == (2, 12, 256)
+    # bias actually reaches the projection: perturbing it changes output
+    with torch.no_grad():
+        m.layers[0].bqkv.fill_(0.5)
+    h2 = m(ids, mask)
+    assert (h - h2).abs().max().item() > 1e-4
