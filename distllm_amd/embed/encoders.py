@@ -39,6 +39,8 @@ def resolve_preset(name_or_path: str) -> str:
     if name_or_path in _HF_NAME_MAP:
         return _HF_NAME_MAP[name_or_path]
     low = name_or_path.lower()
+    if "roberta" in low:
+        return "roberta-base"
     if "pubmedbert" in low or "bert" in low:
         return "pubmedbert-base"
     if "esm2" in low:

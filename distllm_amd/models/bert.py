@@ -125,8 +125,14 @@ class BertEncoder(nn.Module):
                 and token_type_ids is None
                 and os.environ.get("DISTLLM_FORCE_VARLEN")):
             return self._forward_packed(ids, attention_mask)
-        pos = torch.arange(S, device=ids.device)
-        x = self.word_emb[ids] + self.pos_emb[pos][None, :, :]
+        if self.dims.position_style == "roberta":
+            pad = self.dims.pad_token_id
+            nz = (ids != pad).int()
+            pos2d = (nz.cumsum(1) * nz + pad).long()
+            x = self.word_emb[ids] + self.pos_emb[pos2d]
+        else:
+            pos = torch.arange(S, device=ids.device)
+            x = self.word_emb[ids] + self.pos_emb[pos][None, :, :]
         if token_type_ids is not None:
             x = x + self.type_emb[token_type_ids.long()]
         else:
@@ -156,7 +162,11 @@ class BertEncoder(nn.Module):
         lens = attention_mask.int().sum(1)
         cu = torch.zeros(B + 1, dtype=torch.int32, device=ids.device)
         cu[1:] = lens.cumsum(0)
-        pos = (attention_mask.int().cumsum(1) - 1).clamp(min=0)[maskb]
+        if d.position_style == "roberta":
+            nz = (ids != d.pad_token_id).int()
+            pos = (nz.cumsum(1) * nz + d.pad_token_id)[maskb]
+        else:
+            pos = (attention_mask.int().cumsum(1) - 1).clamp(min=0)[maskb]
         bmap = ops.varlen_block_map(cu)
         x = (self.word_emb[ids[maskb]] + self.pos_emb[pos.long()]
              + self.type_emb[0])

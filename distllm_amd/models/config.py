@@ -2,11 +2,13 @@
 
 Covers the reference's encoder zoo plus the
 decoder family used for generation (K15):
-  - BERT-class bidirectional encoders (S-PubMedBERT, pubmedbert, bert-base)
+  - BERT-class bidirectional encoders (S-PubMedBERT, pubmedbert,
+    bert-base) and RoBERTa-class (offset position ids)
   - ESM2 protein encoders (rotary, no token types)
   - ESMC protein encoders
   - Mistral-class decoder embedders (SFR-Embedding-Mistral)
   - Llama-class causal LMs for generation
+  - Qwen2-class causal LMs (llama family + qkv bias)
 Dims are from the public model cards; weights are random-init here
 (no network) or loadable from local safetensors.
 """
@@ -34,6 +36,10 @@ class ModelDims(BaseModel):
     gelu: Literal["tanh", "erf"] = "erf"
     tie_word_embeddings: bool = False
     attention_bias: bool = False      # Qwen2: bias on q/k/v only
+    # RoBERTa-family: positions = cumsum(ids != pad) + pad (HF
+    # create_position_ids_from_input_ids), vs BERT's arange
+    position_style: Literal["absolute", "roberta"] = "absolute"
+    pad_token_id: int = 0
 
     @property
     def head_dim(self) -> int:
@@ -55,6 +61,14 @@ PRESETS: dict[str, ModelDims] = {
     "bert-tiny": ModelDims(  # test-size
         arch="bert", vocab_size=512, hidden_size=128, num_layers=2,
         num_heads=2, intermediate_size=256, max_position_embeddings=128,
+    ),
+    # RoBERTa-base (FacebookAI/roberta-base): BERT-family encoder with
+    # offset position ids and a single token type
+    "roberta-base": ModelDims(
+        arch="bert", vocab_size=50265, hidden_size=768, num_layers=12,
+        num_heads=12, intermediate_size=3072,
+        max_position_embeddings=514, layer_norm_eps=1e-5, gelu="erf",
+        type_vocab_size=1, position_style="roberta", pad_token_id=1,
     ),
     # ESM2 sizes (facebook/esm2_t33_650M_UR50D etc.)
     "esm2-8m": ModelDims(

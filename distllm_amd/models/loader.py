@@ -32,8 +32,14 @@ def _load_safetensors_dir(path: Path) -> dict[str, torch.Tensor]:
 
 
 def _map_bert(src: dict, model) -> dict:
+    """HF BertModel / RobertaModel layouts (identical key structure
+    under different prefixes)."""
     out = {}
-    p = "bert." if any(k.startswith("bert.") for k in src) else ""
+    p = ""
+    for cand in ("bert.", "roberta."):
+        if any(k.startswith(cand) for k in src):
+            p = cand
+            break
     out["word_emb"] = src[f"{p}embeddings.word_embeddings.weight"]
     out["pos_emb"] = src[f"{p}embeddings.position_embeddings.weight"]
     out["type_emb"] = src[f"{p}embeddings.token_type_embeddings.weight"]

@@ -240,3 +240,36 @@ def test_qwen2_matches_hf(tmp_path):
         theirs = hf(input_ids=ids, attention_mask=mask).last_hidden_state
         got = ours(ids, mask)
     _assert_valid_close(got, theirs, [24, 24], "qwen2")
+
+
+def test_roberta_matches_hf(tmp_path):
+    """RoBERTa = BERT family + offset position ids derived from the
+    pad token (HF create_position_ids_from_input_ids)."""
+    from transformers import RobertaConfig, RobertaModel
+
+    dims = ModelDims(
+        arch="bert", vocab_size=256, hidden_size=64, num_layers=3,
+        num_heads=4, intermediate_size=128, max_position_embeddings=66,
+        layer_norm_eps=1e-5, gelu="erf", type_vocab_size=1,
+        position_style="roberta", pad_token_id=1,
+    )
+    cfg = RobertaConfig(
+        vocab_size=256, hidden_size=64, num_hidden_layers=3,
+        num_attention_heads=4, intermediate_size=128,
+        max_position_embeddings=66, type_vocab_size=1,
+        layer_norm_eps=1e-5, hidden_act="gelu", pad_token_id=1,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(6)
+    hf = RobertaModel(cfg, add_pooling_layer=False).eval()
+    hf.save_pretrained(tmp_path / "hf", safe_serialization=True)
+
+    ours = build_model(dims, device="cpu", dtype=torch.float32)
+    load_weights(ours, tmp_path / "hf")
+
+    ids, mask, lens = _mixed_batch(256, avoid=(1,), seed=29)
+    ids[mask == 0] = 1  # pad token drives RoBERTa position ids
+    with torch.no_grad():
+        theirs = hf(input_ids=ids, attention_mask=mask).last_hidden_state
+        got = ours(ids, mask)
+    _assert_valid_close(got, theirs, lens, "roberta")
