@@ -502,3 +502,37 @@ def test_bert_packed_matches_padded(device, monkeypatch):
     m = mask.bool()
     err = (got.float().cpu()[m] - ref.float()[m]).abs().max().item()
     assert err < 0.08, f"packed bert vs padded fp32: max err {err}"
+
+
+def test_roberta_packed_matches_padded(device, monkeypatch):
+    """RoBERTa position convention (offset ids from pad token) through
+    the packed varlen path vs the padded fp32 CPU forward."""
+    from distllm_amd.models import build_model
+    from distllm_amd.models.config import ModelDims
+
+    dims = ModelDims(
+        arch="bert", vocab_size=512, hidden_size=128, num_layers=2,
+        num_heads=2, intermediate_size=256, max_position_embeddings=130,
+        layer_norm_eps=1e-5, type_vocab_size=1,
+        position_style="roberta", pad_token_id=1,
+    )
+    monkeypatch.setenv("DISTLLM_FORCE_VARLEN", "1")
+    torch.manual_seed(62)
+    cpu = build_model(dims, device="cpu", dtype=torch.float32)
+    gpu = build_model(dims, device=device, dtype=torch.bfloat16)
+    gpu.load_state_dict(
+        {k: v.to(device, torch.bfloat16) for k, v in
+         cpu.state_dict().items()})
+    B, S = 4, 64
+    ids = torch.randint(4, 500, (B, S))
+    mask = torch.ones(B, S, dtype=torch.long)
+    for r, L in enumerate([64, 33, 9, 50]):
+        mask[r, L:] = 0
+        ids[r, L:] = 1  # roberta pad token
+    monkeypatch.delenv("DISTLLM_FORCE_VARLEN")
+    ref = cpu(ids, mask)
+    monkeypatch.setenv("DISTLLM_FORCE_VARLEN", "1")
+    got = gpu(ids.to(device), mask.to(device))
+    m = mask.bool()
+    err = (got.float().cpu()[m] - ref.float()[m]).abs().max().item()
+    assert err < 0.08, f"packed roberta vs padded fp32: max err {err}"
