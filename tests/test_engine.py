@@ -231,3 +231,23 @@ def test_native_generator_beam_path():
                          "max_batch_size": 2, "max_model_len": 128})
     out = gen.generate(["hello world"])
     assert len(out) == 1 and isinstance(out[0], str)
+
+
+def test_block_manager_invariants():
+    """Allocator invariants: block 0 reserved, exhaustion raises,
+    release returns capacity, blocks_for rounds up."""
+    from distllm_amd.engine.engine import BlockManager
+
+    bm = BlockManager(num_blocks=8, blk=16)
+    assert 0 not in bm.free  # dummy graph-slot block stays reserved
+    assert bm.blocks_for(1) == 1 and bm.blocks_for(16) == 1
+    assert bm.blocks_for(17) == 2 and bm.blocks_for(0) == 0
+    assert bm.can_alloc(7) and not bm.can_alloc(8)
+    a = bm.alloc(3)
+    b = bm.alloc(4)
+    assert len(set(a) | set(b)) == 7  # disjoint
+    import pytest as _p
+    with _p.raises(RuntimeError):
+        bm.alloc(1)
+    bm.release(a)
+    assert bm.can_alloc(3) and not bm.can_alloc(4)
