@@ -209,6 +209,28 @@ def evaluate(
     run_eval_suite(EvalSuiteConfig.from_yaml(config))
 
 
+@app.command()
+def launch(
+    config: Path = typer.Option(..., help="HpcLaunchConfig YAML"),
+    command: str = typer.Option(
+        ..., help="module command to run under torchrun, e.g. "
+        "'-m distllm_amd.distributed_embedding --config embed.yaml'"),
+    output: Path = typer.Option(
+        None, help="write the batch script here (default: stdout)"),
+) -> None:
+    """Render a Slurm/PBS batch script for a multi-node torchrun job
+    (the reference's Parsl HPC platform role)."""
+    from .parallel.launcher import HpcLaunchConfig, render_batch_script
+
+    script = render_batch_script(HpcLaunchConfig.from_yaml(config),
+                                 command)
+    if output is not None:
+        output.write_text(script)
+        typer.echo(f"wrote {output}")
+    else:
+        typer.echo(script)
+
+
 def main() -> None:
     app()
 

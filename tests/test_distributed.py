@@ -134,3 +134,54 @@ def test_distributed_generation_two_ranks(tmp_path):
 
     total = sum(len(load_from_disk(str(s))) for s in shards)
     assert total == 4
+
+
+def test_hpc_batch_script_rendering(tmp_path):
+    """Slurm and PBS scripts carry the scheduler directives and a
+    c10d-rendezvous torchrun line (reference Parsl platform parity)."""
+    from distllm_amd.parallel.launcher import (
+        HpcLaunchConfig,
+        render_batch_script,
+    )
+
+    cmd = "-m distllm_amd.distributed_embedding --config e.yaml"
+    s = render_batch_script(HpcLaunchConfig(
+        scheduler="slurm", account="proj1", queue="gpu", num_nodes=4,
+        walltime="02:00:00", scheduler_options="#SBATCH --exclusive",
+        worker_init="module load rocm"), cmd)
+    assert "#SBATCH --account=proj1" in s
+    assert "#SBATCH --partition=gpu" in s
+    assert "#SBATCH --nodes=4" in s
+    assert "#SBATCH --exclusive" in s
+    assert "module load rocm" in s
+    assert "--nnodes=4 --nproc-per-node=8" in s
+    assert "--rdzv-backend=c10d" in s
+    assert "scontrol show hostnames" in s
+    assert "HSA_ENABLE_IPC_MODE_LEGACY=0" in s
+    assert s.startswith("#!/bin/bash")
+
+    p = render_batch_script(HpcLaunchConfig(
+        scheduler="pbs", account="proj2", queue="prod", num_nodes=2,
+        gpus_per_node=4), cmd)
+    assert "#PBS -A proj2" in p and "#PBS -q prod" in p
+    assert "#PBS -l select=2:ngpus=4" in p
+    assert "mpiexec -n 2 -ppn 1" in p
+    assert "--nproc-per-node=4" in p
+    assert 'head -n1 "$PBS_NODEFILE"' in p
+
+    # CLI round-trip through YAML
+    import yaml
+
+    cfgp = tmp_path / "hpc.yaml"
+    cfgp.write_text(yaml.safe_dump(
+        {"scheduler": "slurm", "account": "a", "num_nodes": 2}))
+    from typer.testing import CliRunner
+
+    from distllm_amd.cli import app
+
+    out = tmp_path / "job.sh"
+    r = CliRunner().invoke(app, [
+        "launch", "--config", str(cfgp), "--command", cmd,
+        "--output", str(out)])
+    assert r.exit_code == 0, r.output
+    assert "#SBATCH --account=a" in out.read_text()
