@@ -188,6 +188,9 @@ def test_example_configs_parse():
     ChatAppConfig(**yaml.safe_load(open("examples/chat.yaml")))
     MCQAConfig(**yaml.safe_load(open("examples/mcqa.yaml")))
     EvalSuiteConfig(**yaml.safe_load(open("examples/evaluate.yaml")))
+    from distllm_amd.parallel.launcher import HpcLaunchConfig
+
+    HpcLaunchConfig(**yaml.safe_load(open("examples/hpc.yaml")))
 
 
 def test_cli_generate(tmp_path):
