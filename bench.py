@@ -43,6 +43,9 @@ def main() -> None:
     ap.add_argument("--pooler", type=str, default="mean",
                     choices=["mean", "last_token"])
     ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--weight-dtype", type=str, default="bf16",
+                    choices=["bf16", "fp8_e4m3"],
+                    help="generate mode: W8A16 fp8 decode weights")
     args = ap.parse_args()
 
     if args.mode == "generate":
@@ -168,7 +171,8 @@ def bench_generate(args) -> None:
     model = build_model(model_name, device=device, dtype=dtype)
     B, P = args.batch_size, args.prompt_len
     engine = LLMEngine(model, max_batch_size=B,
-                       max_model_len=P + args.steps + args.warmup + 8)
+                       max_model_len=P + args.steps + args.warmup + 8,
+                       weight_dtype=args.weight_dtype)
     params = SamplingParams(temperature=0.5, min_p=0.1,
                             max_tokens=args.steps + args.warmup + 4,
                             stop_token_ids=())
@@ -219,7 +223,8 @@ def bench_generate(args) -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if have_gpu else "fp32",
+            "dtype": ("bf16" if args.weight_dtype == "bf16"
+                      else "bf16+fp8w") if have_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": model_name,

@@ -85,9 +85,21 @@ class LLMEngine:
                  kv_memory_fraction: float = 0.6,
                  num_blocks: int | None = None,
                  use_graphs: bool | None = None,
-                 kv_cache_dtype: str = "bf16"):
+                 kv_cache_dtype: str = "bf16",
+                 weight_dtype: str = "bf16"):
         self.model = model
         d = model.dims
+        # optional W8A16 decode weights (OCP fp8-e4m3, per-channel
+        # scales): the B<=8 decode step is weight-stream bound, so fp8
+        # halves its HBM bytes (models/mistral.py enable_fp8_decode;
+        # prefill and larger batches keep bf16)
+        if weight_dtype not in ("bf16", "fp8_e4m3"):
+            raise ValueError(f"weight_dtype {weight_dtype!r}")
+        self.weight_dtype = weight_dtype
+        if weight_dtype == "fp8_e4m3":
+            if not str(model.device).startswith("cuda"):
+                raise ValueError("fp8_e4m3 weights require a GPU")
+            model.enable_fp8_decode()
         self.device = model.device
         self.max_batch_size = max_batch_size
         self.max_model_len = min(max_model_len, d.max_position_embeddings)

@@ -110,6 +110,25 @@ class DecoderModel(nn.Module):
         self.reset_parameters()
         self.to(dtype)
         self._rope_cache: tuple[torch.Tensor, torch.Tensor] | None = None
+        self._fp8w: list[dict] | None = None
+
+    @torch.no_grad()
+    def enable_fp8_decode(self) -> None:
+        """Opt-in W8A16 decode: quantize the projection weights to OCP
+        fp8-e4m3 (per-output-channel scales) for the small-batch GEMV
+        path. Prefill and B > 8 decode keep the bf16 weights — the fp8
+        copies add ~0.5 byte/param. Idempotent."""
+        if self._fp8w is not None:
+            return
+        fp8 = []
+        for layer in self.layers:
+            fp8.append({
+                "wqkv": ops.quantize_fp8_rows(layer.wqkv),
+                "wo": ops.quantize_fp8_rows(layer.wo),
+                "wgup": ops.quantize_fp8_rows(layer.wgup),
+                "wdown": ops.quantize_fp8_rows(layer.wdown),
+            })
+        self._fp8w = fp8
 
     @torch.no_grad()
     def reset_parameters(self):
@@ -244,12 +263,20 @@ class DecoderModel(nn.Module):
         x = self.tok_emb[tokens.long()].unsqueeze(1)  # [B,1,H]
         pos = (ctx_lens - 1).int()
         slot = pos
+        # W8A16 small-batch path: fp8 weight GEMV (enable_fp8_decode)
+        fp8 = (self._fp8w if self._fp8w is not None and B <= 8
+               and x.is_cuda else None)
         h = ops.rms_norm(x, self.layers[0].in_norm, eps=d.rms_norm_eps)
         for li, layer in enumerate(self.layers):
             # ONE fused qkv GEMM + ONE rope (q|k roped, v copied):
             # 5 launches -> 2 on the per-token critical path
-            qkv = ops.linear(h, layer.wqkv,
-                             layer.bqkv).view(B, 1, nh + 2 * kvh, hd)
+            if fp8 is not None:
+                f = fp8[li]
+                qkv = ops.gemv_fp8(h[:, 0], *f["wqkv"], layer.bqkv)
+                qkv = qkv.view(B, 1, nh + 2 * kvh, hd)
+            else:
+                qkv = ops.linear(h, layer.wqkv,
+                                 layer.bqkv).view(B, 1, nh + 2 * kvh, hd)
             qkv = ops.rope_apply(qkv, cs, sn, pos, rope_heads=nh + kvh)
             q = qkv[:, 0, :nh]
             k = qkv[:, 0, nh: nh + kvh].contiguous()
@@ -259,13 +286,23 @@ class DecoderModel(nn.Module):
             ctx = ops.paged_attn(q.contiguous(), kcaches[li],
                                  vcaches[li], block_tables, ctx_lens,
                                  scale=1.0 / math.sqrt(hd))
-            attn = self._allreduce(
-                ops.linear(ctx.reshape(B, 1, nh * hd), layer.wo))
+            if fp8 is not None:
+                attn = self._allreduce(ops.gemv_fp8(
+                    ctx.reshape(B, nh * hd), *f["wo"])).unsqueeze(1)
+            else:
+                attn = self._allreduce(
+                    ops.linear(ctx.reshape(B, 1, nh * hd), layer.wo))
             h2, x = ops.norm_fwd_sum(attn, x, layer.post_norm,
                                      eps=d.rms_norm_eps, rms=True)
-            down = self._allreduce(
-                ops.linear(ops.gateup_swiglu(h2, layer.wgup),
-                           layer.wdown))
+            if fp8 is not None:
+                gu = ops.gemv_fp8(h2[:, 0], *f["wgup"])
+                inter = ops._hip_ops.swiglu_packed(gu)
+                down = self._allreduce(
+                    ops.gemv_fp8(inter, *f["wdown"])).unsqueeze(1)
+            else:
+                down = self._allreduce(
+                    ops.linear(ops.gateup_swiglu(h2, layer.wgup),
+                               layer.wdown))
             next_w = (self.layers[li + 1].in_norm
                       if li + 1 < len(self.layers) else self.final_norm)
             h, x = ops.norm_fwd_sum(down, x, next_w,

@@ -417,6 +417,26 @@ def gateup_swiglu(x, wgup, bias=None):
     return out.reshape(*orig_shape[:-1], I)
 
 
+def quantize_fp8_rows(w: torch.Tensor):
+    """Per-output-channel OCP fp8-e4m3 weight quantization for the
+    W8A16 decode GEMV: returns (wq uint8 [N, K], scale f32 [N]) with
+    scale = absmax/448 so dequant is wq * scale per row."""
+    wf = w.detach().float()
+    scale = (wf.abs().amax(1) / 448.0).clamp_min(1e-12)
+    q = (wf / scale[:, None]).clamp(-448.0, 448.0)
+    q = q.to(torch.float8_e4m3fn).view(torch.uint8)
+    return q.contiguous(), scale.contiguous()
+
+
+def gemv_fp8(x, wq, scale, bias=None):
+    """out = x @ (dequant(wq) * scale)^T (+bias). x bf16 [M<=8, K],
+    wq uint8 fp8-e4m3 [N, K], scale f32 [N]. One wave per output
+    column; the fp32 dot runs on the quantized values, the channel
+    scale applies after — see csrc/gemv_fp8.hip."""
+    _require_hip()
+    return _hip_ops.gemv_fp8(x.contiguous(), wq, scale, bias)
+
+
 rope_tables = torch_ref.rope_tables
 
 

@@ -57,6 +57,8 @@ void launch_hamming_score(const void*, const void*, float*, int, long, int,
                           hipStream_t);
 void launch_ivf_score(const void*, const int*, const int*, const float*,
                       float*, int, int, int, int, hipStream_t);
+void launch_gemv_fp8(const void*, const void*, const float*, const void*,
+                     void*, int, int, long, hipStream_t);
 void launch_sample(const float*, const float*, int*, int, int, float, float,
                    hipStream_t);
 void launch_paged_attn_split(const void*, const void*, const void*,
@@ -446,6 +448,33 @@ torch::Tensor ivf_score(torch::Tensor slabs, torch::Tensor lens,
   return out;
 }
 
+torch::Tensor gemv_fp8(torch::Tensor x, torch::Tensor wq,
+                       torch::Tensor scale,
+                       c10::optional<torch::Tensor> bias) {
+  CHECK_DEV(x); CHECK_CONT(x); CHECK_BF16(x);
+  CHECK_DEV(wq); CHECK_CONT(wq);
+  TORCH_CHECK(wq.scalar_type() == at::kByte, "wq must be uint8 (fp8)");
+  CHECK_DEV(scale); CHECK_CONT(scale); CHECK_F32(scale);
+  const int M = x.size(0), N = wq.size(0);
+  const long K = wq.size(1);
+  TORCH_CHECK(x.size(1) == K, "K mismatch");
+  TORCH_CHECK(M >= 1 && M <= 8, "gemv_fp8 serves M <= 8 decode rows");
+  TORCH_CHECK(K % 16 == 0, "K must be a multiple of 16");
+  TORCH_CHECK(scale.size(0) == N, "scale must be per output channel");
+  const void* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_DEV(bias.value()); CHECK_CONT(bias.value());
+    CHECK_BF16(bias.value());
+    TORCH_CHECK(bias->size(0) == N, "bias shape");
+    bptr = bias->data_ptr();
+  }
+  auto out = torch::empty({M, N}, x.options());
+  launch_gemv_fp8(x.data_ptr(), wq.data_ptr(),
+                  scale.data_ptr<float>(), bptr, out.data_ptr(), M, N,
+                  K, cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("blt_linear", &blt_linear,
         "hipBLASLt linear with fused bias/GELU/swish epilogue");
@@ -470,6 +499,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_chunked", &topk_chunked);
   m.def("hamming_score", &hamming_score);
   m.def("ivf_score", &ivf_score);
+  m.def("gemv_fp8", &gemv_fp8, py::arg("x"), py::arg("wq"),
+        py::arg("scale"), py::arg("bias") = py::none());
   m.def("sample_tokens", &sample_tokens);
   m.def("paged_attn", &paged_attn);
   m.def("cache_put", &cache_put);
