@@ -53,8 +53,12 @@ class NativeGeneratorConfig(BaseConfig):
     # rank (data-parallel across ranks) in this round:
     tensor_parallel_size: int = 1
     trust_remote_code: bool = True
+    # reference `quantization: true` meant NF4 capacity quantization;
+    # here it opts into the CDNA4 equivalent: W8A16 fp8-e4m3 decode
+    # weights (weight_dtype below)
     quantization: bool = False
     kv_cache_dtype: str = "bf16"  # or "fp8_e4m3" (CDNA4 OCP fp8)
+    weight_dtype: str = "bf16"    # or "fp8_e4m3" (W8A16 decode GEMV)
 
 
 class NativeGenerator:
@@ -96,10 +100,20 @@ class NativeGenerator:
             arch=dims.arch, vocab_size=dims.vocab_size,
             model_max_length=dims.max_position_embeddings,
         )
+        wdt = config.weight_dtype
+        if config.quantization and wdt == "bf16":
+            if str(self.model.device).startswith("cuda"):
+                wdt = "fp8_e4m3"
+            else:  # reference-YAML compat field; inert off-GPU
+                import warnings
+
+                warnings.warn("quantization requested but no GPU; "
+                              "keeping bf16 weights")
         self.engine = LLMEngine(self.model,
                                 max_batch_size=config.max_batch_size,
                                 max_model_len=config.max_model_len,
-                                kv_cache_dtype=config.kv_cache_dtype)
+                                kv_cache_dtype=config.kv_cache_dtype,
+                                weight_dtype=wdt)
         # vLLM-style capture-at-init: pre-record every pow2 decode
         # graph bucket so batch drain never captures inside a timed /
         # serving window (measured ~0.2s per first-use capture)
