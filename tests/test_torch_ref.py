@@ -268,3 +268,23 @@ def test_varlen_block_map_edge_cases():
     assert m([0, 65]) == ([0, 0], [0, 1])
     # zero-length sequence in the middle is skipped entirely
     assert m([0, 64, 64, 200]) == ([0, 2, 2, 2], [0, 0, 1, 2])
+
+
+def test_quantize_fp8_rows_cpu():
+    """Per-channel fp8-e4m3 weight quantization: dequant error bounded
+    by the e4m3 step (~6% relative at per-row absmax scaling), exact
+    zeros, and scale shape/positivity."""
+    from distllm_amd.ops import quantize_fp8_rows
+
+    torch.manual_seed(80)
+    w = torch.randn(16, 64, dtype=torch.bfloat16) * 3.0
+    w[0] = 0.0
+    wq, sc = quantize_fp8_rows(w)
+    assert wq.dtype == torch.uint8 and wq.shape == w.shape
+    assert sc.dtype == torch.float32 and sc.shape == (16,)
+    assert (sc > 0).all()
+    wd = wq.view(torch.float8_e4m3fn).float() * sc[:, None]
+    assert torch.equal(wd[0], torch.zeros(64))
+    rel = (wd[1:] - w[1:].float()).abs().max(1).values / \
+        w[1:].float().abs().max(1).values
+    assert rel.max().item() < 0.08
