@@ -143,3 +143,26 @@ def test_top_p_sampling_seeded_and_min_p():
     _E._rng = torch.Generator().manual_seed(1)
     toks = LLMEngine._sample_top_p(_E(), spiked, params2)
     assert (toks == 5).all()
+
+
+def test_quantization_flag_cpu_warns_and_stays_bf16():
+    """Reference YAMLs with `quantization: true` (NF4 upstream) map to
+    the fp8-e4m3 weight path on GPU; off-GPU the flag warns and the
+    generator still works in bf16."""
+    import warnings
+
+    from distllm_amd.generate.generators import (
+        NativeGenerator,
+        NativeGeneratorConfig,
+    )
+
+    cfg = NativeGeneratorConfig(name="native", model="llama-tiny",
+                                quantization=True, max_tokens=4,
+                                max_model_len=128)
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        gen = NativeGenerator(cfg)
+        assert any("quantization" in str(x.message) for x in w)
+    assert gen.engine.weight_dtype == "bf16"
+    out = gen.generate(["hello there"])
+    assert len(out) == 1
