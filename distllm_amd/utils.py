@@ -11,6 +11,7 @@ from __future__ import annotations
 import json
 import re
 import time
+from dataclasses import dataclass
 from pathlib import Path
 from typing import Any, Iterator, TypeVar
 
@@ -72,7 +73,14 @@ def iter_batches(data: list[T], batch_size: int) -> Iterator[list[T]]:
 # Timing
 # ---------------------------------------------------------------------------
 
-class TimeStats(BaseModel):
+@dataclass
+class TimeStats:
+    """A dataclass (not a pydantic model) so the reference's analysis
+    idiom `pd.DataFrame(time_stats)` expands rows into the
+    tags/elapsed_s/start_unix/end_unix columns (reference
+    tests/distllm_test.py:44-48; pydantic-v2 models iterate as
+    (name, value) tuples and would scramble that)."""
+
     tags: list[str]
     elapsed_s: float
     start_unix: float

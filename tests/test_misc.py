@@ -261,3 +261,29 @@ def test_qwen2_preset_and_cpu_forward():
         m.layers[0].bqkv.fill_(0.5)
     h2 = m(ids, mask)
     assert (h - h2).abs().max().item() > 1e-4
+
+
+def test_timer_stats_pandas_contract(capsys, tmp_path):
+    """Reference analysis idiom (reference tests/distllm_test.py:44-48):
+    Timer output parsed from captured stdout converts to a DataFrame
+    with the tags/elapsed_s/start_unix/end_unix columns; Path tags are
+    accepted."""
+    from pathlib import Path as _P
+
+    import pandas as pd
+
+    from distllm_amd.utils import TimeLogger, Timer
+
+    with Timer("func1", _P("/my/path1")):
+        pass
+    with Timer("func2", _P("/my/path2")):
+        pass
+    log = tmp_path / "time.log"
+    log.write_text(capsys.readouterr().out)
+    stats = TimeLogger().parse_logs(log)
+    assert stats[0].tags == ["func1", "/my/path1"]
+    assert stats[1].tags == ["func2", "/my/path2"]
+    df = pd.DataFrame(stats)
+    assert len(df) == 2
+    assert list(df.keys()) == ["tags", "elapsed_s", "start_unix",
+                               "end_unix"]
