@@ -219,11 +219,24 @@ def launch(
         None, help="write the batch script here (default: stdout)"),
 ) -> None:
     """Render a Slurm/PBS batch script for a multi-node torchrun job
-    (the reference's Parsl HPC platform role)."""
-    from .parallel.launcher import HpcLaunchConfig, render_batch_script
+    (the reference's Parsl HPC platform role). Accepts either an
+    HpcLaunchConfig YAML or a full pipeline YAML whose compute_config
+    names a reference HPC platform (polaris/leonardo/slurm/pbs)."""
+    import yaml as _yaml
 
-    script = render_batch_script(HpcLaunchConfig.from_yaml(config),
-                                 command)
+    from .parallel.launcher import HpcLaunchConfig, render_batch_script
+    from .parallel.runtime import ComputeConfig
+
+    raw = _yaml.safe_load(Path(config).read_text())
+    if isinstance(raw, dict) and "compute_config" in raw:
+        lc = ComputeConfig(**raw["compute_config"]).to_launch_config()
+        if lc is None:
+            raise typer.BadParameter(
+                "compute_config names an in-process platform; use "
+                "name: polaris/leonardo/slurm/pbs for batch launch")
+    else:
+        lc = HpcLaunchConfig(**raw)
+    script = render_batch_script(lc, command)
     if output is not None:
         output.write_text(script)
         typer.echo(f"wrote {output}")

@@ -32,11 +32,41 @@ class ComputeConfig(BaseConfig):
     name "local": current process only (reference LocalConfig);
     name "node":  one rank per GPU via torch.distributed (reference
     WorkstationConfig's one-worker-per-GPU pinning, minus Parsl).
+    The reference's HPC platform names (reference parsl.py: polaris =
+    PBSPro, leonardo = Slurm) are accepted drop-in: in-process they
+    behave like "node" (torchrun provides the ranks), and
+    to_launch_config() maps the scheduler fields onto the batch-script
+    renderer (`distllm launch`).
     """
 
-    name: Literal["local", "node"] = "local"
+    name: Literal["local", "node", "workstation", "slurm", "pbs",
+                  "polaris", "leonardo"] = "local"
     ranks_per_node: int = 8
     backend: str = ""  # "" -> nccl if cuda else gloo
+    # reference HPC platform fields (parsl.py:106-250)
+    num_nodes: int = 1
+    account: str = ""
+    queue: str = ""
+    walltime: str = "01:00:00"
+    worker_init: str = ""
+    scheduler_options: str = ""
+
+    def to_launch_config(self):
+        """HpcLaunchConfig for batch-script rendering, or None for the
+        in-process platforms."""
+        sched = {"slurm": "slurm", "leonardo": "slurm",
+                 "pbs": "pbs", "polaris": "pbs"}.get(self.name)
+        if sched is None:
+            return None
+        from .launcher import HpcLaunchConfig
+
+        return HpcLaunchConfig(
+            scheduler=sched, account=self.account, queue=self.queue,
+            walltime=self.walltime, num_nodes=self.num_nodes,
+            gpus_per_node=self.ranks_per_node,
+            worker_init=self.worker_init,
+            scheduler_options=self.scheduler_options,
+        )
 
 
 def init_distributed(backend: str = "") -> tuple[int, int]:

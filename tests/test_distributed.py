@@ -185,3 +185,55 @@ def test_hpc_batch_script_rendering(tmp_path):
         "--output", str(out)])
     assert r.exit_code == 0, r.output
     assert "#SBATCH --account=a" in out.read_text()
+
+
+def test_reference_platform_compute_configs(tmp_path):
+    """Reference-shaped compute_config blocks (parsl.py platform
+    fields) parse drop-in and map onto the batch-script renderer."""
+    import yaml
+
+    from distllm_amd.parallel.launcher import render_batch_script
+    from distllm_amd.parallel.runtime import ComputeConfig
+
+    # polaris (PBSPro) shape from reference examples/embed/*.yaml
+    cc = ComputeConfig(
+        name="polaris", num_nodes=16,
+        worker_init="module load rocm",
+        scheduler_options="#PBS -l filesystems=home:eagle:grand",
+        account="FoundEpidem", queue="demand", walltime="01:00:00")
+    lc = cc.to_launch_config()
+    s = render_batch_script(lc, "-m distllm_amd.distributed_embedding")
+    assert "#PBS -A FoundEpidem" in s and "#PBS -q demand" in s
+    assert "#PBS -l filesystems=home:eagle:grand" in s
+    assert "select=16" in s and "module load rocm" in s
+
+    # leonardo (Slurm) maps to sbatch
+    lc2 = ComputeConfig(name="leonardo", account="x",
+                        walltime="00:30:00").to_launch_config()
+    assert lc2.scheduler == "slurm"
+    # in-process platforms render nothing
+    assert ComputeConfig(name="node").to_launch_config() is None
+
+    # `distllm launch` consumes a full pipeline YAML's compute_config
+    from typer.testing import CliRunner
+
+    from distllm_amd.cli import app
+
+    p = tmp_path / "embed.yaml"
+    p.write_text(yaml.safe_dump({
+        "input_dir": "/x", "output_dir": "/y",
+        "dataset_config": {"name": "jsonl"},
+        "encoder_config": {"name": "auto"},
+        "pooler_config": {"name": "mean"},
+        "embedder_config": {"name": "full_sequence"},
+        "writer_config": {"name": "huggingface"},
+        "compute_config": {"name": "polaris", "account": "proj",
+                           "queue": "prod", "num_nodes": 2},
+    }))
+    out = tmp_path / "job.sh"
+    r = CliRunner().invoke(app, [
+        "launch", "--config", str(p),
+        "--command", "-m distllm_amd.distributed_embedding "
+        f"--config {p}", "--output", str(out)])
+    assert r.exit_code == 0, r.output
+    assert "#PBS -A proj" in out.read_text()
