@@ -27,13 +27,35 @@ class NativeGeneratorConfig(BaseConfig):
     @model_validator(mode="before")
     @classmethod
     def _huggingface_defaults(cls, data: Any) -> Any:
-        # Reference HuggingFaceGenerator defaults differ from vLLM's:
-        # num_beams=10, top_p=.95, do_sample=True
-        # (huggingface_backend.py:41-44). YAMLs naming the huggingface
-        # backend and relying on defaults get those semantics here too.
-        if isinstance(data, dict) and data.get("name") == "huggingface":
-            data.setdefault("num_beams", 10)
-            data.setdefault("top_p", 0.95)
+        # Drop-in acceptance of BOTH reference backend dialects:
+        #  - vllm_backend.py: `llm_name` names the model
+        #  - huggingface_backend.py: `pretrained_model_name_or_path`,
+        #    `batch_size`, `do_sample` (False -> greedy), plus inert
+        #    runtime knobs (half_precision/eval_mode/compile_model)
+        #    the native engine supersedes (bf16, inference-only,
+        #    hand-fused).
+        if isinstance(data, dict):
+            for alias in ("llm_name", "pretrained_model_name_or_path"):
+                if alias in data and "model" not in data:
+                    data["model"] = data.pop(alias)
+                else:
+                    data.pop(alias, None)
+            if "batch_size" in data:
+                data.setdefault("max_batch_size",
+                                data.pop("batch_size"))
+            if data.pop("do_sample", True) is False:
+                data.setdefault("temperature", 0.0)
+            for inert in ("half_precision", "eval_mode",
+                          "compile_model"):
+                data.pop(inert, None)
+            # Reference HuggingFaceGenerator defaults differ from
+            # vLLM's: num_beams=10, top_p=.95, do_sample=True
+            # (huggingface_backend.py:41-44). YAMLs naming the
+            # huggingface backend and relying on defaults get those
+            # semantics here too.
+            if data.get("name") == "huggingface":
+                data.setdefault("num_beams", 10)
+                data.setdefault("top_p", 0.95)
         return data
     model: str = "llama-tiny"
     weights_path: str = ""

@@ -166,3 +166,26 @@ def test_quantization_flag_cpu_warns_and_stays_bf16():
     assert gen.engine.weight_dtype == "bf16"
     out = gen.generate(["hello there"])
     assert len(out) == 1
+
+
+def test_reference_generator_dialects_parse():
+    """Reference vllm/huggingface backend YAML fields parse drop-in
+    (vllm_backend.py llm_name; huggingface_backend.py
+    pretrained_model_name_or_path/batch_size/do_sample + inert runtime
+    knobs)."""
+    from distllm_amd.generate.generators import NativeGeneratorConfig
+
+    v = NativeGeneratorConfig(
+        name="vllm", llm_name="mistralai/Mistral-7B-Instruct-v0.2",
+        top_p=0.95, tensor_parallel_size=1, trust_remote_code=True)
+    assert v.model == "mistralai/Mistral-7B-Instruct-v0.2"
+
+    h = NativeGeneratorConfig(
+        name="huggingface",
+        pretrained_model_name_or_path="meta-llama/Llama-3-8b",
+        half_precision=True, eval_mode=True, compile_model=False,
+        quantization=False, do_sample=False, batch_size=4)
+    assert h.model == "meta-llama/Llama-3-8b"
+    assert h.max_batch_size == 4
+    assert h.temperature == 0.0  # do_sample: false -> greedy
+    assert h.num_beams == 10     # reference hf defaults still apply
