@@ -13,7 +13,7 @@ from __future__ import annotations
 from typing import Any, Literal, Union
 
 import torch
-from pydantic import Field
+from pydantic import Field, model_validator
 
 from ..models import build_model, get_dims
 from ..models.config import PRESETS, ModelDims
@@ -94,8 +94,22 @@ def _default_device() -> str:
     return "cuda" if torch.cuda.is_available() else "cpu"
 
 
+def _ref_tokenizer_alias(data):
+    """Reference YAMLs in the wild spell the tokenizer field
+    `tokenizer_name` (reference examples/embed/*.yaml; its pydantic
+    accepted unknown keys). Map it onto tokenizer_path."""
+    if isinstance(data, dict):
+        if "tokenizer_name" in data and not data.get("tokenizer_path"):
+            data["tokenizer_path"] = data.pop("tokenizer_name")
+        else:
+            data.pop("tokenizer_name", None)
+    return data
+
+
 class AutoEncoderConfig(BaseConfig):
     """Text encoders (BERT-class, Mistral-class)."""
+
+    _ref_alias = model_validator(mode="before")(_ref_tokenizer_alias)
 
     name: Literal["auto"] = "auto"
     pretrained_model_name_or_path: str = "pubmedbert-base"
@@ -110,20 +124,28 @@ class AutoEncoderConfig(BaseConfig):
 
 
 class Esm2EncoderConfig(BaseConfig):
+    _ref_alias = model_validator(mode="before")(_ref_tokenizer_alias)
+
     name: Literal["esm2"] = "esm2"
     pretrained_model_name_or_path: str = "esm2-650m"
     tokenizer_path: str = ""
     weights_path: str = ""
     half_precision: bool = True
+    eval_mode: bool = True       # reference esm2.py:27 (always eval)
+    compile_model: bool = False  # reference esm2.py:29 (hand-fused)
     faesm: bool = False  # flash attention is always on here
 
 
 class EsmCambrianEncoderConfig(BaseConfig):
+    _ref_alias = model_validator(mode="before")(_ref_tokenizer_alias)
+
     name: Literal["esmc"] = "esmc"
     pretrained_model_name_or_path: str = "esmc-300m"
     tokenizer_path: str = ""
     weights_path: str = ""
     half_precision: bool = True
+    eval_mode: bool = True
+    compile_model: bool = False
 
 
 def _build(cfg) -> _NativeEncoder:

@@ -189,3 +189,23 @@ def test_reference_generator_dialects_parse():
     assert h.max_batch_size == 4
     assert h.temperature == 0.0  # do_sample: false -> greedy
     assert h.num_beams == 10     # reference hf defaults still apply
+
+
+def test_reference_encoder_dialects_parse():
+    """Reference encoder YAML shapes (incl. the shipped examples'
+    tokenizer_name spelling and esm eval_mode/compile_model) parse
+    drop-in."""
+    from distllm_amd.embed.encoders import (
+        AutoEncoderConfig,
+        Esm2EncoderConfig,
+        EsmCambrianEncoderConfig,
+    )
+
+    a = AutoEncoderConfig(
+        pretrained_model_name_or_path="some/checkpoint-63423",
+        tokenizer_name="bert-base-uncased", quantization=False)
+    assert a.tokenizer_path == "bert-base-uncased"
+    e = Esm2EncoderConfig(eval_mode=True, compile_model=False,
+                          faesm=True)
+    assert e.half_precision
+    EsmCambrianEncoderConfig(eval_mode=True, compile_model=False)
