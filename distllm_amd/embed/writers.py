@@ -19,6 +19,10 @@ from .embedders import EmbedderResult
 
 class HuggingFaceWriterConfig(BaseConfig):
     name: Literal["huggingface"] = "huggingface"
+    # reference writers/huggingface.py:43 — dataset-writing worker
+    # count; accepted for YAML drop-in (shards here are written by
+    # their owning rank, one process each)
+    num_proc: int | None = None
 
 
 class HuggingFaceWriter:
