@@ -51,7 +51,7 @@ class HuggingFaceGenWriter:
 
 class AMPJsonlWriterConfig(BaseConfig):
     name: Literal["amp_json"] = "amp_json"
-    base_name: str = "amp_results"
+    base_name: str = "question_set"  # reference amp_json.py:21 default
 
 
 class AMPJsonlWriter:
