@@ -43,6 +43,7 @@ class ModelConfig(BaseConfig):
     base_url: str = ""
     model: str = "distllm-amd"
     api_key_env: str = "OPENAI_API_KEY"
+    api_key: str = ""  # literal key (reference v2 generator_settings)
     temperature: float = 0.0
     max_tokens: int = 512
     boot_local: bool = False
@@ -60,6 +61,7 @@ class GraderConfig(BaseConfig):
     base_url: str = ""
     model: str = "distllm-amd"
     api_key_env: str = "OPENAI_API_KEY"
+    api_key: str = ""
     temperature: float = 0.0
     max_tokens: int = 256
 
@@ -84,6 +86,82 @@ class OutputConfig(BaseConfig):
     checkpoint_dir: Path = Path("mcqa_checkpoints")
 
 
+def _load_model_server(shortname: str, config_file: str | Path) -> dict:
+    """Resolve a grader/model shortname from an argonium-style
+    model_servers.yaml (reference v2:603-634: servers: [{shortname,
+    openai_api_base, openai_api_key, openai_model}, ...])."""
+    import yaml as _yaml
+
+    data = _yaml.safe_load(Path(config_file).read_text()) or {}
+    for server in data.get("servers", []):
+        if server.get("shortname") == shortname:
+            return server
+    avail = [s.get("shortname") for s in data.get("servers", [])]
+    raise ValueError(f"model {shortname!r} not found in "
+                     f"{config_file}; available: {avail}")
+
+
+def translate_v2_config(raw: dict, base_dir: Path | None = None) -> dict:
+    """Map the reference MCQA v2/v3 YAML tree (reference
+    rag_argonium_score_parallel_v2.py config schema: nested
+    model.generator/generator_settings, rag.enabled/retrieval_top_k,
+    processing.*, output.output_directory+output_prefix) onto
+    MCQAConfig's tree. Native-shaped dicts pass through untouched."""
+    model = raw.get("model")
+    if not (isinstance(model, dict)
+            and ("generator" in model or "generator_settings" in model)):
+        return raw
+    gs = model.get("generator_settings", {}) or {}
+    base_url = gs.get("base_url", "")
+    if not base_url and gs.get("server"):
+        base_url = f"http://{gs['server']}:{gs.get('port', 8000)}/v1"
+    out: dict = {"questions_file": raw.get("questions_file", "")}
+    out["model"] = {
+        "base_url": base_url,
+        "model": gs.get("model", "distllm-amd"),
+        "api_key": str(gs.get("api_key", "") or ""),
+        "temperature": gs.get("temperature", 0.0),
+        "max_tokens": gs.get("max_tokens", 512),
+    }
+    grader: dict = {}
+    if model.get("grader_shortname"):
+        cf = Path(model.get("model_config_file", "model_servers.yaml"))
+        if not cf.is_absolute() and base_dir is not None:
+            cf = base_dir / cf
+        if cf.exists():
+            srv = _load_model_server(model["grader_shortname"], cf)
+            grader = {
+                "base_url": srv.get("openai_api_base", ""),
+                "model": srv.get("openai_model",
+                                 model["grader_shortname"]),
+                "api_key": str(srv.get("openai_api_key", "") or ""),
+            }
+        else:
+            grader = {"model": model["grader_shortname"]}
+    out["grader"] = grader or dict(out["model"])
+    rag = raw.get("rag", {}) or {}
+    out["rag"] = {
+        "use_rag": bool(rag.get("enabled", False)),
+        "retriever_config": rag.get("retriever_config"),
+        "top_k": rag.get("retrieval_top_k", 5),
+        "score_threshold": rag.get("retrieval_score_threshold", 0.0),
+    }
+    proc = raw.get("processing", {}) or {}
+    out["processing"] = {
+        "parallel_workers": proc.get("parallel_workers", 4),
+        "random_selection": proc.get("random_selection") or 0,
+        "seed": proc.get("random_seed") or 42,
+    }
+    o = raw.get("output", {}) or {}
+    outdir = Path(o.get("output_directory", "."))
+    prefix = o.get("output_prefix", "mcqa")
+    out["output"] = {
+        "results_file": outdir / f"{prefix}_results.json",
+        "checkpoint_dir": outdir / f"{prefix}_checkpoints",
+    }
+    return out
+
+
 class MCQAConfig(BaseConfig):
     questions_file: Path
     model: ModelConfig
@@ -91,6 +169,14 @@ class MCQAConfig(BaseConfig):
     rag: RagSettings = RagSettings()
     processing: ProcessingConfig = ProcessingConfig()
     output: OutputConfig = OutputConfig()
+
+    @classmethod
+    def from_yaml(cls, path) -> "MCQAConfig":
+        import yaml as _yaml
+
+        raw = _yaml.safe_load(Path(path).read_text()) or {}
+        raw = translate_v2_config(raw, base_dir=Path(path).parent)
+        return cls(**raw)
 
 
 # ---------------------------------------------------------------------------
@@ -192,13 +278,14 @@ class LocalServer:
 
 class _ChatClient:
     def __init__(self, base_url: str, model: str, api_key_env: str,
-                 temperature: float, max_tokens: int):
+                 temperature: float, max_tokens: int,
+                 api_key: str = ""):
         import os
 
         import httpx
 
         headers = {}
-        key = os.environ.get(api_key_env, "")
+        key = api_key or os.environ.get(api_key_env, "")
         if key:
             headers["Authorization"] = f"Bearer {key}"
         self.client = httpx.Client(base_url=base_url, headers=headers,
@@ -398,12 +485,14 @@ def run_mcqa(config: MCQAConfig) -> dict:
         client = _ChatClient(base_url, config.model.model,
                              config.model.api_key_env,
                              config.model.temperature,
-                             config.model.max_tokens)
+                             config.model.max_tokens,
+                             api_key=config.model.api_key)
         grader = _ChatClient(config.grader.base_url or base_url,
                              config.grader.model,
                              config.grader.api_key_env,
                              config.grader.temperature,
-                             config.grader.max_tokens)
+                             config.grader.max_tokens,
+                             api_key=config.grader.api_key)
         retriever = None
         if config.rag.use_rag and config.rag.retriever_config:
             from ..rag.search import RetrieverConfig

@@ -270,3 +270,59 @@ def test_chat_target_dispatch(tmp_path, monkeypatch):
     import pytest as _p
     with _p.raises(ValueError, match="Unknown generator config"):
         translate_target_config({"_target_": "NopeConfig"})
+
+
+def test_mcqa_v2_yaml_translation(tmp_path):
+    """Reference MCQA v2/v3-shaped YAML (nested generator_settings,
+    rag.enabled, processing/output blocks, model_servers.yaml grader
+    lookup) loads drop-in through MCQAConfig.from_yaml."""
+    import yaml
+
+    from distllm_amd.mcqa.harness import MCQAConfig
+
+    (tmp_path / "model_servers.yaml").write_text(yaml.safe_dump({
+        "servers": [{"shortname": "gpt-4.1",
+                     "openai_api_base": "http://g:1/v1",
+                     "openai_api_key": "k2",
+                     "openai_model": "gpt-4.1-mini"}]}))
+    cfg_path = tmp_path / "mcqa_v2.yaml"
+    cfg_path.write_text(yaml.safe_dump({
+        "questions_file": "/q.json",
+        "model": {
+            "generator": {"generator_type": "vllm"},
+            "generator_settings": {
+                "server": "rbdgx2", "model": "meta-llama/L70",
+                "port": 9999, "api_key": "CELS",
+                "temperature": 0.0, "max_tokens": 2048},
+            "grader_shortname": "gpt-4.1",
+            "model_config_file": "model_servers.yaml"},
+        "rag": {"enabled": True, "rag_config_file": None,
+                "retriever_config": None, "use_context_field": False,
+                "retrieval_top_k": 5,
+                "retrieval_score_threshold": 0.0,
+                "chunk_logging_enabled": True},
+        "processing": {"parallel_workers": 4, "question_format": "mc",
+                       "verbose": True, "random_selection": 100,
+                       "random_seed": None},
+        "output": {"save_incorrect": True, "output_directory": "/tmp/o",
+                   "output_prefix": "run1"},
+    }))
+    cfg = MCQAConfig.from_yaml(cfg_path)
+    assert cfg.model.base_url == "http://rbdgx2:9999/v1"
+    assert cfg.model.api_key == "CELS"
+    assert cfg.model.max_tokens == 2048
+    assert cfg.grader.base_url == "http://g:1/v1"
+    assert cfg.grader.model == "gpt-4.1-mini"
+    assert cfg.grader.api_key == "k2"
+    assert cfg.rag.use_rag and cfg.rag.top_k == 5
+    assert cfg.processing.random_selection == 100
+    assert str(cfg.output.results_file) == "/tmp/o/run1_results.json"
+
+    # native-shaped YAML passes through unchanged
+    native = tmp_path / "native.yaml"
+    native.write_text(yaml.safe_dump({
+        "questions_file": "/q.json",
+        "model": {"base_url": "http://x/v1"},
+        "grader": {"base_url": "http://y/v1"}}))
+    n = MCQAConfig.from_yaml(native)
+    assert n.model.base_url == "http://x/v1"
