@@ -77,10 +77,16 @@ def translate_target_config(gen: dict[str, Any]) -> dict[str, Any]:
       ArgoGeneratorConfig    {model, base_url, api_key, ...}
       OpenAIAPIGeneratorConfig {model, api_key, base_url?, ...}
 
-    all map to the OpenAI-compatible HTTP client; dicts without
-    `_target_` pass through unchanged."""
+    all map to the OpenAI-compatible HTTP client. Bare
+    server/port-shaped dicts (the reference chat.py
+    VLLMGeneratorSettings blocks in examples/chat/*.yaml) translate the
+    same way; dicts already naming one of our backends pass through
+    unchanged."""
     if "_target_" not in gen:
-        return gen
+        if "name" not in gen and "server" in gen and "port" in gen:
+            gen = {**gen, "_target_": "VLLMGeneratorConfig"}
+        else:
+            return gen
     gen = dict(gen)
     target = gen.pop("_target_")
     common = {
@@ -123,6 +129,18 @@ class ChatAppConfig(BaseConfig):
         with open(path) as fh:
             data = yaml.safe_load(fh)
         data = substitute_env(data or {})
+        # reference chat YAML dialect (examples/chat/*.yaml): nested
+        # rag_configs block + save_conversation_path
+        if isinstance(data.get("rag_configs"), dict):
+            rc = data.pop("rag_configs")
+            data.setdefault("generator_config",
+                            rc.get("generator_config", {}))
+            if rc.get("retriever_config") is not None:
+                data.setdefault("retriever_config",
+                                rc["retriever_config"])
+        if "save_conversation_path" in data:
+            data.setdefault("transcript_dir",
+                            data.pop("save_conversation_path"))
         if isinstance(data.get("generator_config"), dict):
             data["generator_config"] = translate_target_config(
                 data["generator_config"])

@@ -326,3 +326,29 @@ def test_mcqa_v2_yaml_translation(tmp_path):
         "grader": {"base_url": "http://y/v1"}}))
     n = MCQAConfig.from_yaml(native)
     assert n.model.base_url == "http://x/v1"
+
+
+def test_chat_reference_yaml_dialect(tmp_path):
+    """Reference examples/chat/*.yaml shape (nested rag_configs block,
+    bare server/port generator settings, save_conversation_path) loads
+    drop-in through ChatAppConfig.from_yaml."""
+    import yaml
+
+    from distllm_amd.chat import ChatAppConfig
+
+    p = tmp_path / "chat.yaml"
+    p.write_text(yaml.safe_dump({
+        "rag_configs": {
+            "generator_config": {
+                "server": "rbdgx2", "port": 9999,
+                "model": "meta-llama/L70", "api_key": "CELS"},
+            "retriever_config": None,
+            "verbose": False},
+        "save_conversation_path": str(tmp_path / "conv"),
+    }))
+    cfg = ChatAppConfig.from_yaml(p)
+    g = cfg.generator_config
+    assert g["name"] == "openai_http"
+    assert g["base_url"] == "http://rbdgx2:9999/v1"
+    assert g["api_key"] == "CELS" and g["model"] == "meta-llama/L70"
+    assert str(cfg.transcript_dir) == str(tmp_path / "conv")
