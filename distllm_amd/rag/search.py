@@ -59,9 +59,19 @@ class HipIndexConfig(BaseConfig):
     name: str = Field("hip_index",
                       description="'hip_index' (alias: 'faiss_index_v2')")
     dataset_dir: Path
+    # defaults beside the dataset — reference FaissIndexV1 YAMLs
+    # (rag/search.py V1, examples/evaluate/*.yaml) give only
+    # dataset_dir
     faiss_index_path: Path = Field(
-        description="index cache path (packed codes + meta, .npz)"
+        Path(""),
+        description="index cache path (packed codes + meta, .npz); "
+        "empty -> <dataset_dir>/hip_index.npz",
     )
+
+    def model_post_init(self, _ctx) -> None:
+        if str(self.faiss_index_path) in ("", "."):
+            self.faiss_index_path = Path(self.dataset_dir) / \
+                "hip_index.npz"
     dataset_chunk_paths: list[Path] | None = None
     precision: str = "float32"
     search_algorithm: str = "exact"

@@ -209,3 +209,12 @@ def test_reference_encoder_dialects_parse():
                           faesm=True)
     assert e.half_precision
     EsmCambrianEncoderConfig(eval_mode=True, compile_model=False)
+
+
+def test_faiss_v1_yaml_shape_parses():
+    """Reference FaissIndexV1 YAML blocks (examples/evaluate/*.yaml)
+    give only dataset_dir; the index path defaults alongside it."""
+    from distllm_amd.rag.search import HipIndexConfig
+
+    c = HipIndexConfig(name="faiss_index_v1", dataset_dir="/data/m")
+    assert str(c.faiss_index_path) == "/data/m/hip_index.npz"
