@@ -180,3 +180,35 @@ def test_run_eval_suite_end_to_end(tmp_path):
     assert rows[0]["task"] == "sciq" and 0.0 <= rows[0]["accuracy"] <= 1.0
     saved = json.loads((tmp_path / "results.json").read_text())
     assert saved == rows
+
+
+def test_reference_evaluate_yaml_shape_parses():
+    """A reference examples/evaluate/*.yaml-shaped config (V1 faiss
+    block, vllm llm_name dialect, nested retriever tree) validates."""
+    import yaml
+
+    from distllm_amd.rag.evaluate import EvalSuiteConfig
+
+    raw = yaml.safe_load("""
+rag_configs:
+  - generator_config:
+      name: vllm
+      llm_name: some/model
+      top_p: 0.95
+      tensor_parallel_size: 8
+    retriever_config:
+      faiss_config:
+        name: faiss_index_v1
+        dataset_dir: /data/merged
+      encoder_config:
+        name: auto
+        pretrained_model_name_or_path: pritamdeka/S-PubMedBert-MS-MARCO
+      pooler_config:
+        name: mean
+tasks: [sciq, pubmedqa, litqa, protein_interaction_qa, protein_function_qa]
+download_dir: /data/benchmarks
+""")
+    c = EvalSuiteConfig(**raw)
+    assert len(c.rag_configs) == 1
+    fc = c.rag_configs[0].retriever_config.faiss_config
+    assert str(fc.faiss_index_path).endswith("hip_index.npz")
