@@ -18,29 +18,51 @@ app = typer.Typer(no_args_is_help=True, pretty_exceptions_enable=False)
 
 @app.command()
 def embed(
-    encoder_name: str = typer.Option("auto", help="auto|esm2|esmc"),
+    # both the reference's snake_case spellings + short aliases
+    # (reference cli.py:15-128) and the dash-case typer default work
+    encoder_name: str = typer.Option(
+        "auto", "--encoder-name", "--encoder_name", "-mn",
+        help="auto|esm2|esmc"),
     pretrained_model_name_or_path: str = typer.Option(
-        "pubmedbert-base", help="model preset or local checkpoint"
-    ),
-    data_path: Path = typer.Option(..., help="input file or directory"),
-    data_extension: str = typer.Option("jsonl"),
-    output_path: Path = typer.Option(...),
+        "pubmedbert-base", "--pretrained-model-name-or-path",
+        "--pretrained_model_name_or_path", "-m",
+        help="model preset or local checkpoint"),
+    data_path: Path = typer.Option(
+        ..., "--data-path", "--data_path", "-d",
+        help="input file or directory"),
+    data_extension: str = typer.Option(
+        "jsonl", "--data-extension", "--data_extension", "-de"),
+    output_path: Path = typer.Option(
+        ..., "--output-path", "--output_path", "-o"),
     dataset_name: str = typer.Option(
-        "jsonl_chunk",
-        help="jsonl|jsonl_chunk|fasta|sequence_per_line|huggingface",
-    ),
-    batch_size: int = typer.Option(8),
-    chunk_batch_size: int = typer.Option(8),
-    buffer_size: int = typer.Option(1),
-    pooler_name: str = typer.Option("mean", help="mean|last_token"),
+        "jsonl_chunk", "--dataset-name", "--dataset_name", "-dn",
+        help="jsonl|jsonl_chunk|fasta|sequence_per_line|huggingface"),
+    batch_size: int = typer.Option(
+        8, "--batch-size", "--batch_size", "-b"),
+    chunk_batch_size: int = typer.Option(
+        8, "--chunk-batch-size", "--chunk_batch_size", "-cb"),
+    buffer_size: int = typer.Option(
+        1, "--buffer-size", "--buffer_size", "-bs"),
+    pooler_name: str = typer.Option(
+        "mean", "--pooler-name", "--pooler_name", "-pn",
+        help="mean|last_token"),
     embedder_name: str = typer.Option(
-        "semantic_chunk", help="full_sequence|semantic_chunk"
-    ),
-    writer_name: str = typer.Option("huggingface", help="huggingface|numpy"),
-    half_precision: bool = typer.Option(True),
-    eval_mode: bool = typer.Option(True),
-    compile_model: bool = typer.Option(False),
-    quantization: bool = typer.Option(False),
+        "semantic_chunk", "--embedder-name", "--embedder_name", "-en",
+        help="full_sequence|semantic_chunk"),
+    writer_name: str = typer.Option(
+        "huggingface", "--writer-name", "--writer_name", "-wn",
+        help="huggingface|numpy"),
+    half_precision: bool = typer.Option(
+        True, "--half-precision/--no-half-precision",
+        "--half_precision/--no-half_precision", "-hp"),
+    eval_mode: bool = typer.Option(
+        True, "--eval-mode/--no-eval-mode",
+        "--eval_mode/--no-eval_mode", "-em"),
+    compile_model: bool = typer.Option(
+        False, "--compile-model/--no-compile-model",
+        "--compile_model/--no-compile_model", "-cm"),
+    quantization: bool = typer.Option(
+        False, "--quantization/--no-quantization", "-q"),
 ) -> None:
     """Embed a file/directory serially in-process (single GPU)."""
     from .distributed_embedding import embedding_worker
@@ -72,10 +94,15 @@ def embed(
 
 @app.command()
 def merge(
-    writer_name: str = typer.Option("huggingface"),
-    num_proc: int = typer.Option(1),
-    dataset_dir: Path = typer.Option(..., help="dir of dataset shards"),
-    output_dir: Path = typer.Option(...),
+    writer_name: str = typer.Option(
+        "huggingface", "--writer-name", "--writer_name", "-wn"),
+    num_proc: int = typer.Option(
+        1, "--num-proc", "--num_proc", "-np"),
+    dataset_dir: Path = typer.Option(
+        ..., "--dataset-dir", "--dataset_dir", "-d",
+        help="dir of dataset shards"),
+    output_dir: Path = typer.Option(
+        ..., "--output-dir", "--output_dir", "-o"),
 ) -> None:
     """Merge per-file dataset shards into one dataset."""
     from .embed import get_writer
@@ -88,19 +115,36 @@ def merge(
 
 @app.command()
 def generate(
-    input_dir: Path = typer.Option(...),
-    output_dir: Path = typer.Option(...),
-    prompt_name: str = typer.Option("question_chunk"),
-    reader_name: str = typer.Option("huggingface"),
-    writer_name: str = typer.Option("huggingface"),
-    generator_name: str = typer.Option("native", help="native|vllm|openai_http"),
-    llm_name: str = typer.Option("llama3-8b"),
-    temperature: float = typer.Option(0.5),
-    min_p: float = typer.Option(0.1),
-    top_p: float = typer.Option(0.0),
-    max_tokens: int = typer.Option(2000),
-    batch_size: int = typer.Option(64),
-    quantization: bool = typer.Option(False),
+    input_dir: Path = typer.Option(
+        ..., "--input-dir", "--input_dir", "-i"),
+    output_dir: Path = typer.Option(
+        ..., "--output-dir", "--output_dir", "-o"),
+    prompt_name: str = typer.Option(
+        "question_chunk", "--prompt-name", "--prompt_name", "-pn"),
+    reader_name: str = typer.Option(
+        "huggingface", "--reader-name", "--reader_name", "-rn"),
+    writer_name: str = typer.Option(
+        "huggingface", "--writer-name", "--writer_name", "-wn"),
+    generator_name: str = typer.Option(
+        "native", "--generator-name", "--generator_name", "-gn",
+        help="native|vllm|openai_http"),
+    llm_name: str = typer.Option(
+        "llama3-8b", "--llm-name", "--llm_name", "-vmn"),
+    temperature: float = typer.Option(
+        0.5, "--temperature", "-t"),
+    min_p: float = typer.Option(0.1, "--min-p", "--min_p", "-mp"),
+    top_p: float = typer.Option(0.0, "--top-p", "--top_p", "-tp"),
+    max_tokens: int = typer.Option(
+        2000, "--max-tokens", "--max_tokens", "-mt"),
+    use_beam_search: bool = typer.Option(
+        False, "--use-beam-search/--no-use-beam-search",
+        "--use_beam_search/--no-use_beam_search",
+        help="beam search (reference vLLM flag; num_beams=10)"),
+    batch_size: int = typer.Option(
+        64, "--batch-size", "--batch_size", "-b"),
+    quantization: bool = typer.Option(
+        False, "--quantization/--no-quantization", "-q",
+        help="W8A16 fp8-e4m3 decode weights on GPU"),
 ) -> None:
     """Generate over a directory of inputs serially in-process."""
     from .distributed_generation import generate_worker
@@ -113,7 +157,10 @@ def generate(
         "top_p": top_p,
         "max_tokens": max_tokens,
         "max_batch_size": batch_size,
+        "quantization": quantization,
     }
+    if use_beam_search:
+        generator_kwargs["num_beams"] = 10
     if generator_name in ("openai_http", "argo"):
         generator_kwargs = {"name": generator_name, "model": llm_name,
                             "temperature": temperature,
@@ -139,15 +186,29 @@ def generate(
 
 @app.command()
 def tokenize(
-    input_dir: Path = typer.Option(...),
-    output_dir: Path = typer.Option(...),
-    text_field: str = typer.Option("text"),
-    tokenizer_name: str = typer.Option("pubmedbert-base"),
-    save_labels: bool = typer.Option(False),
+    input_dir: Path = typer.Option(
+        ..., "--input-dir", "--input_dir", "-i"),
+    output_dir: Path = typer.Option(
+        ..., "--output-dir", "--output_dir", "-o"),
+    text_field: str = typer.Option(
+        "text", "--text-field", "--text_field", "-tf"),
+    tokenizer_name: str = typer.Option(
+        "pubmedbert-base", "--tokenizer-name", "--tokenizer_name",
+        "-tn"),
+    dotenv_path: Path = typer.Option(
+        None, "--dotenv-path", "--dotenv_path", "-dp",
+        help="env file loaded before tokenizing (reference flag)"),
+    save_labels: bool = typer.Option(
+        False, "--save-labels/--no-save-labels",
+        "--save_labels/--no-save_labels", "-sl"),
 ) -> None:
     """Tokenize jsonl files into HF datasets."""
     from .distributed_tokenization import TokenizationConfig, main as tmain
 
+    if dotenv_path is not None:
+        from .generate.generators import ApiGenerator
+
+        ApiGenerator._load_dotenv(str(dotenv_path))
     tmain(TokenizationConfig(
         input_dir=input_dir, output_dir=output_dir, text_field=text_field,
         tokenizer_name=tokenizer_name, save_labels=save_labels,
@@ -156,9 +217,13 @@ def tokenize(
 
 @app.command()
 def chunk_fasta_file(
-    input_file: Path = typer.Option(...),
-    output_dir: Path = typer.Option(...),
-    num_chunks: int = typer.Option(8),
+    input_file: Path = typer.Option(
+        ..., "--input-file", "--input_file", "-i"),
+    output_dir: Path = typer.Option(
+        ..., "--output-dir", "--output_dir", "-o"),
+    num_chunks: int = typer.Option(
+        8, "--num-chunks", "--chunk_size", "-c",
+        help="number of smaller files (reference --chunk_size)"),
 ) -> None:
     """Split one fasta file into N chunk files."""
     from .embed.datasets import read_fasta, write_fasta

@@ -352,3 +352,25 @@ def test_chat_reference_yaml_dialect(tmp_path):
     assert g["base_url"] == "http://rbdgx2:9999/v1"
     assert g["api_key"] == "CELS" and g["model"] == "meta-llama/L70"
     assert str(cfg.transcript_dir) == str(tmp_path / "conv")
+
+
+def test_cli_reference_flag_spellings(tmp_path):
+    """The reference CLI's snake_case flags and short aliases
+    (reference cli.py: --encoder_name/-mn, --chunk_size/-c, ...) work
+    alongside the dash-case spellings."""
+    from typer.testing import CliRunner
+
+    from distllm_amd.cli import app
+
+    fa = tmp_path / "x.fasta"
+    fa.write_text(">a\nMKV\n>b\nMLL\n>c\nMAA\n>d\nMCC\n")
+    out = tmp_path / "chunks"
+    r = CliRunner().invoke(app, [
+        "chunk-fasta-file", "--input_file", str(fa),
+        "--output_dir", str(out), "--chunk_size", "2"])
+    assert r.exit_code == 0, r.output
+    assert len(list(out.glob("*.fasta"))) == 2
+    out2 = tmp_path / "chunks2"
+    r2 = CliRunner().invoke(app, [
+        "chunk-fasta-file", "-i", str(fa), "-o", str(out2), "-c", "4"])
+    assert r2.exit_code == 0, r2.output
