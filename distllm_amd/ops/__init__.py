@@ -351,7 +351,8 @@ def varlen_block_map(cu_seqlens: torch.Tensor, qblk: int = 128):
     starts = torch.zeros(len(lens), dtype=torch.int64)
     if len(lens) > 1:
         starts[1:] = nqb.cumsum(0)[:-1]
-    blk_qb = torch.arange(total, dtype=torch.int64) -         torch.repeat_interleave(starts, nqb)
+    blk_qb = (torch.arange(total, dtype=torch.int64)
+              - torch.repeat_interleave(starts, nqb))
     dev = cu_seqlens.device
     return blk_seq.int().to(dev), blk_qb.int().to(dev)
 

@@ -67,11 +67,6 @@ class HipIndexConfig(BaseConfig):
         description="index cache path (packed codes + meta, .npz); "
         "empty -> <dataset_dir>/hip_index.npz",
     )
-
-    def model_post_init(self, _ctx) -> None:
-        if str(self.faiss_index_path) in ("", "."):
-            self.faiss_index_path = Path(self.dataset_dir) / \
-                "hip_index.npz"
     dataset_chunk_paths: list[Path] | None = None
     precision: str = "float32"
     search_algorithm: str = "exact"
@@ -84,6 +79,11 @@ class HipIndexConfig(BaseConfig):
     nlist: int = 0    # ivf: clusters (0 -> ~sqrt(N))
     nprobe: int = 32  # ivf: chunks probed per query (recall@10 0.94
     # at 10M rows/nprobe 32, 0.97 at 64 — profiles/RESULTS.md)
+
+    def model_post_init(self, _ctx) -> None:
+        if str(self.faiss_index_path) in ("", "."):
+            self.faiss_index_path = (Path(self.dataset_dir)
+                                     / "hip_index.npz")
 
 
 class HipFlatIndex:
