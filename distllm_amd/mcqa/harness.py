@@ -140,9 +140,18 @@ def translate_v2_config(raw: dict, base_dir: Path | None = None) -> dict:
             grader = {"model": model["grader_shortname"]}
     out["grader"] = grader or dict(out["model"])
     rag = raw.get("rag", {}) or {}
+    retr = rag.get("retriever_config")
+    if retr is None and rag.get("rag_config_file"):
+        import yaml as _yaml
+
+        rf = Path(rag["rag_config_file"])
+        if not rf.is_absolute() and base_dir is not None:
+            rf = base_dir / rf
+        if rf.exists():
+            retr = _yaml.safe_load(rf.read_text())
     out["rag"] = {
         "use_rag": bool(rag.get("enabled", False)),
-        "retriever_config": rag.get("retriever_config"),
+        "retriever_config": retr,
         "top_k": rag.get("retrieval_top_k", 5),
         "score_threshold": rag.get("retrieval_score_threshold", 0.0),
     }
