@@ -98,11 +98,12 @@ def _ref_tokenizer_alias(data):
     """Reference YAMLs in the wild spell the tokenizer field
     `tokenizer_name` (reference examples/embed/*.yaml; its pydantic
     accepted unknown keys). Map it onto tokenizer_path."""
-    if isinstance(data, dict):
-        if "tokenizer_name" in data and not data.get("tokenizer_path"):
+    if isinstance(data, dict) and "tokenizer_name" in data:
+        data = dict(data)  # never mutate the caller's dict
+        if not data.get("tokenizer_path"):
             data["tokenizer_path"] = data.pop("tokenizer_name")
         else:
-            data.pop("tokenizer_name", None)
+            data.pop("tokenizer_name")
     return data
 
 

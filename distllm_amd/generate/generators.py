@@ -35,6 +35,7 @@ class NativeGeneratorConfig(BaseConfig):
         #    the native engine supersedes (bf16, inference-only,
         #    hand-fused).
         if isinstance(data, dict):
+            data = dict(data)  # never mutate the caller's dict
             for alias in ("llm_name", "pretrained_model_name_or_path"):
                 if alias in data and "model" not in data:
                     data["model"] = data.pop(alias)
