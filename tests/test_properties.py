@@ -102,3 +102,27 @@ def test_config_yaml_roundtrip(tmp_path_factory, name, k, frac):
     back = _Cfg.from_yaml(p)
     assert back.name == name and back.k == k
     assert abs(back.frac - frac) <= 1e-6 * max(1.0, abs(frac))
+
+
+_jsonish = st.recursive(
+    st.none() | st.booleans() | st.integers() | st.floats(
+        allow_nan=False) | st.text(max_size=10),
+    lambda c: st.lists(c, max_size=3) | st.dictionaries(
+        st.text(max_size=8), c, max_size=4),
+    max_leaves=12,
+)
+
+
+@settings(deadline=None, max_examples=150)
+@given(st.dictionaries(st.text(max_size=12), _jsonish, max_size=6))
+def test_config_translators_total_on_junk(data):
+    """The reference-dialect translators are total functions over
+    arbitrary JSON-ish dicts: they either map or pass through, never
+    crash (validation errors belong to pydantic afterwards)."""
+    from distllm_amd.embed.encoders import _ref_tokenizer_alias
+    from distllm_amd.mcqa.harness import translate_v2_config
+
+    out = translate_v2_config(dict(data))
+    assert isinstance(out, dict)
+    out2 = _ref_tokenizer_alias(dict(data))
+    assert isinstance(out2, dict)
