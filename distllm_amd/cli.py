@@ -16,6 +16,19 @@ import typer
 app = typer.Typer(no_args_is_help=True, pretty_exceptions_enable=False)
 
 
+@app.callback(invoke_without_command=True)
+def _main_options(
+    version: bool = typer.Option(
+        False, "--version", "-V", is_eager=True,
+        help="print the distllm-amd version and exit"),
+) -> None:
+    if version:
+        import distllm_amd
+
+        typer.echo(f"distllm-amd {distllm_amd.__version__}")
+        raise typer.Exit()
+
+
 @app.command()
 def embed(
     # both the reference's snake_case spellings + short aliases
