@@ -198,8 +198,8 @@ class HipFlatIndex:
 
     def _build_ivf(self) -> None:
         """Batched IVF: GPU k-means, then clusters split into chunks of
-        at most `cap` rows and stored as PADDED bf16 slabs
-        [nchunks, cap, dim] with per-chunk row->corpus-id maps. Search
+        at most `cap` rows and stored as PADDED fp32 slabs
+        [nchunks, cap, dim_pad4] with per-chunk row->corpus-id maps. Search
         scores every probed chunk with one regular kernel launch (no
         per-query host loop) and top-ks over the padded candidate
         matrix. Replaces the reference's faiss HNSW approximate option
@@ -261,7 +261,7 @@ class HipFlatIndex:
         self.ivf_cap = cap
         self.ivf_nchunks = nch
         self.ivf_centroids = ccent        # [nchunks, dim] chunk means
-        self.ivf_slabs = slabs            # [nchunks, cap, dim] bf16
+        self.ivf_slabs = slabs            # [nchunks, cap, dim_pad4] f32
         self.ivf_idx = idx_pad.to(dev)    # [nchunks, cap] int32, -1 pad
         self.ivf_len = lens.to(dev)       # [nchunks] int32
 
